@@ -1,0 +1,113 @@
+"""LoRA adapters — replaces the reference's peft dependency
+(LoraConfig r=64, alpha=16, target q/k/v/o/gate_up/down + fully-trained
+embed_tokens/lm_head via modules_to_save, GRPO/grpo.py:226-243).
+
+MI355X-first detail: instead of peft's merge_and_unload-to-disk before every
+rollout (grpo_trainer.py:131-140), `merge_for_rollout()` materializes
+W + (alpha/r)·B·A into a cached HBM buffer that the in-process sampler reads
+directly (288 GB leaves room for a merged copy), and `unmerge()` drops it.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LoraConfig:
+    r: int = 64
+    alpha: int = 16
+    dropout: float = 0.0
+    target_modules: tuple = ("q_proj", "k_proj", "v_proj", "o_proj",
+                             "gate_up_proj", "down_proj")
+    # fully-trained modules (reference modules_to_save: embed_tokens, lm_head)
+    modules_to_save: tuple = ("embed_tokens", "lm_head")
+
+
+class LoRALinear(nn.Module):
+    def __init__(self, base: nn.Linear, r: int, alpha: int, dropout: float = 0.0):
+        super().__init__()
+        self.base = base
+        for p in self.base.parameters():
+            p.requires_grad_(False)
+        dtype = base.weight.dtype
+        self.lora_A = nn.Parameter(torch.zeros(r, base.in_features, dtype=dtype))
+        self.lora_B = nn.Parameter(torch.zeros(base.out_features, r, dtype=dtype))
+        nn.init.kaiming_uniform_(self.lora_A, a=5 ** 0.5)
+        self.scaling = alpha / r
+        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self._merged: torch.Tensor | None = None
+
+    @property
+    def in_features(self):
+        return self.base.in_features
+
+    @property
+    def out_features(self):
+        return self.base.out_features
+
+    @property
+    def bias(self):
+        return self.base.bias
+
+    @property
+    def weight(self):
+        return self.base.weight
+
+    def forward(self, x):
+        if self._merged is not None and not torch.is_grad_enabled():
+            return F.linear(x, self._merged, self.base.bias)
+        y = self.base(x)
+        lx = self.dropout(x)
+        return y + F.linear(F.linear(lx, self.lora_A), self.lora_B) * self.scaling
+
+    def merge_for_rollout(self):
+        with torch.no_grad():
+            self._merged = (self.base.weight.float()
+                            + (self.lora_B.float() @ self.lora_A.float()) * self.scaling
+                            ).to(self.base.weight.dtype)
+
+    def unmerge(self):
+        self._merged = None
+
+
+def apply_lora(model: nn.Module, cfg: LoraConfig) -> nn.Module:
+    """Wrap target linears with LoRA; freeze everything except adapters and
+    modules_to_save.  Returns the (mutated) model."""
+    for p in model.parameters():
+        p.requires_grad_(False)
+    for name, module in model.named_modules():
+        for child_name, child in list(module.named_children()):
+            if child_name in cfg.target_modules and isinstance(child, nn.Linear):
+                setattr(module, child_name, LoRALinear(child, cfg.r, cfg.alpha, cfg.dropout))
+    for name, module in model.named_modules():
+        leaf = name.rsplit(".", 1)[-1]
+        if leaf in cfg.modules_to_save:
+            for p in module.parameters(recurse=True):
+                p.requires_grad_(True)
+    return model
+
+
+def merge_for_rollout(model: nn.Module):
+    for m in model.modules():
+        if isinstance(m, LoRALinear):
+            m.merge_for_rollout()
+
+
+def unmerge(model: nn.Module):
+    for m in model.modules():
+        if isinstance(m, LoRALinear):
+            m.unmerge()
+
+
+def lora_state_dict(model: nn.Module, cfg: LoraConfig) -> dict:
+    """Adapter + modules_to_save tensors only (what checkpoints serialize —
+    mirrors peft save_pretrained + modules_to_save, grpo_trainer.py:321-327)."""
+    out = {}
+    for name, p in model.named_parameters():
+        if "lora_A" in name or "lora_B" in name or p.requires_grad:
+            out[name] = p.detach().cpu()
+    return out

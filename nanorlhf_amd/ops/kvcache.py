@@ -1,0 +1,60 @@
+"""Paged KV-cache primitives for the in-process sampler.
+
+Replaces vLLM's paged attention.  The pool is two
+tensors [num_pages, page_size, Hkv, D] (K and V) in bf16, sized against
+288 GB HBM3E; sequences own page lists (block tables).
+
+  * kv_append: scatter freshly-projected K/V rows into their page slots.
+  * paged_attn_decode: one query token per sequence; HIP kernel does
+    online-softmax attention over the sequence's pages, one workgroup per
+    (sequence, kv-head) with the q-heads of that group held in registers
+    (GQA), vectorized bf16 loads of KV lines, fp32 accumulation.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import ext
+
+
+def kv_append(k: torch.Tensor, v: torch.Tensor, slots: torch.Tensor,
+              k_cache: torch.Tensor, v_cache: torch.Tensor) -> None:
+    """k/v: [T, Hkv, D] bf16; slots: [T] int64 flat slot index
+    (page * page_size + offset) into caches viewed as [num_pages*page_size, Hkv, D]."""
+    if k.is_cuda:
+        ext().kv_append(k.contiguous(), v.contiguous(), slots, k_cache, v_cache)
+        return
+    kc = k_cache.view(-1, *k_cache.shape[2:])
+    vc = v_cache.view(-1, *v_cache.shape[2:])
+    kc[slots] = k
+    vc[slots] = v
+
+
+def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+                      block_tables: torch.Tensor, seq_lens: torch.Tensor,
+                      scale: float) -> torch.Tensor:
+    """q: [B, Hq, D] bf16 (one new token per sequence); block_tables [B, P] int32;
+    seq_lens [B] int32 (length INCLUDING the new token, whose K/V are already
+    appended).  Returns [B, Hq, D] bf16."""
+    if q.is_cuda:
+        return ext().paged_attn_decode(q.contiguous(), k_cache, v_cache,
+                                       block_tables.contiguous(), seq_lens.contiguous(),
+                                       float(scale))
+    # CPU reference
+    B, Hq, D = q.shape
+    page = k_cache.shape[1]
+    Hkv = k_cache.shape[2]
+    rep = Hq // Hkv
+    out = torch.empty_like(q, dtype=torch.float32)
+    kc = k_cache.view(-1, Hkv, D).float()
+    vc = v_cache.view(-1, Hkv, D).float()
+    for b in range(B):
+        L = int(seq_lens[b])
+        pages = block_tables[b, : (L + page - 1) // page].long()
+        slots = (pages.unsqueeze(1) * page + torch.arange(page)).reshape(-1)[:L]
+        kk = kc[slots].repeat_interleave(rep, dim=1)  # [L, Hq, D]
+        vv = vc[slots].repeat_interleave(rep, dim=1)
+        att = torch.einsum("hd,lhd->hl", q[b].float(), kk) * scale
+        att = torch.softmax(att, dim=-1)
+        out[b] = torch.einsum("hl,lhd->hd", att, vv)
+    return out.to(q.dtype)

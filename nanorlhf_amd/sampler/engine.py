@@ -1,0 +1,196 @@
+"""In-process generation engine — the vLLM replacement.
+
+Deletes the reference's per-update disk round trip (save LoRA → reload base
+→ merge_and_unload → save merged → boot LLM() → generate → del llm,
+GRPO/grpo_trainer.py:122-166): here the sampler runs the live policy module
+in-place (LoRA pre-merged into an HBM buffer, models/lora.py), with a paged
+KV pool, continuous batching (finished sequences free pages, queued ones are
+admitted), prefill via the varlen MFMA flash kernel and decode via the paged
+LDS-staged attention kernel, temperature/top-p sampling in HIP.
+
+API mirrors the reference's `vllm_generate(N, model, tokenizer, prompts,
+temperature, max_tokens)` contract (grpo_trainer.py:122-166): `generate`
+returns per-prompt responses padded to max_tokens with the pad token.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+
+from .. import ops
+from ..models.qwen2 import AttnContext, CausalLM
+from ..models.lora import merge_for_rollout, unmerge
+from .cache import PagedKVCache, SeqState
+
+
+@dataclass
+class SamplingParams:
+    n: int = 1
+    temperature: float = 1.0
+    top_p: float = 0.95
+    max_tokens: int = 128
+    seed: int = 0
+    stop_token_id: int | None = None  # EOS; None → run to max_tokens
+
+
+class SamplerEngine:
+    def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
+                 max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072):
+        self.model = model
+        self.device = next(model.parameters()).device
+        self.dtype = next(model.parameters()).dtype
+        self.pool = PagedKVCache.for_budget(model.cfg, kv_pool_tokens, page_size,
+                                            device=self.device, dtype=self.dtype)
+        self.max_num_seqs = max_num_seqs
+        self.prefill_chunk_tokens = prefill_chunk_tokens
+
+    # ----------------------------------------------------------------- utils
+    def _slots_for_range(self, seq: SeqState, start: int, end: int) -> list[int]:
+        ps = self.pool.page_size
+        return [seq.slot_of(p, ps) for p in range(start, end)]
+
+    def _sample_from_hidden(self, hidden_last: torch.Tensor,
+                            params: SamplingParams) -> torch.Tensor:
+        logits = self.model.logits(hidden_last)
+        self._sample_step += 1
+        return ops.sample_tokens(logits, params.temperature, params.top_p,
+                                 params.seed, self._sample_step)
+
+    # -------------------------------------------------------------- prefill
+    @torch.no_grad()
+    def _prefill(self, seqs: list[SeqState], params: SamplingParams):
+        """Prefill `seqs` (packed varlen, chunked by token budget), append KV,
+        sample each sequence's first generated token."""
+        i = 0
+        while i < len(seqs):
+            # pack a chunk of sequences up to the token budget
+            chunk: list[SeqState] = []
+            total = 0
+            while i < len(seqs) and (not chunk or total + len(seqs[i]) <= self.prefill_chunk_tokens):
+                total += len(seqs[i])
+                chunk.append(seqs[i])
+                i += 1
+            lens = [len(s) for s in chunk]
+            ids = torch.tensor([t for s in chunk for t in s.tokens], dtype=torch.long,
+                               device=self.device)
+            cu = torch.zeros(len(chunk) + 1, dtype=torch.int32, device=self.device)
+            cu[1:] = torch.cumsum(torch.tensor(lens, dtype=torch.int32, device=self.device), 0)
+            pos = torch.cat([torch.arange(n, device=self.device) for n in lens])
+            slots = torch.tensor([sl for s in chunk for sl in self._slots_for_range(s, 0, len(s))],
+                                 dtype=torch.long, device=self.device)
+            ctx = AttnContext(mode="prefill", positions=pos, cu_seqlens=cu,
+                              max_seqlen=max(lens), kv_caches=self.pool.layers, slots=slots)
+            hidden = self.model(ids, ctx)
+            last_idx = cu[1:].long() - 1
+            tokens = self._sample_from_hidden(hidden[last_idx], params)
+            tok_list = tokens.tolist()
+            for s, t in zip(chunk, tok_list):
+                s.tokens.append(int(t))
+                if params.stop_token_id is not None and int(t) == params.stop_token_id:
+                    s.finished = True
+
+    # --------------------------------------------------------------- decode
+    @torch.no_grad()
+    def _decode_step(self, seqs: list[SeqState], params: SamplingParams):
+        """One decode step over all active sequences."""
+        ps = self.pool.page_size
+        for s in seqs:
+            s.ensure_capacity(self.pool, len(s) + 1)
+        ids = torch.tensor([s.tokens[-1] for s in seqs], dtype=torch.long, device=self.device)
+        pos = torch.tensor([len(s) - 1 for s in seqs], dtype=torch.long, device=self.device)
+        slots = torch.tensor([s.slot_of(len(s) - 1, ps) for s in seqs], dtype=torch.long,
+                             device=self.device)
+        seq_lens = torch.tensor([len(s) for s in seqs], dtype=torch.int32, device=self.device)
+        max_pages = max(len(s.pages) for s in seqs)
+        bt = torch.zeros(len(seqs), max_pages, dtype=torch.int32)
+        for r, s in enumerate(seqs):
+            bt[r, : len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
+        bt = bt.to(self.device)
+        ctx = AttnContext(mode="decode", positions=pos, kv_caches=self.pool.layers,
+                          slots=slots, block_tables=bt, seq_lens=seq_lens)
+        hidden = self.model(ids, ctx)
+        tokens = self._sample_from_hidden(hidden, params)
+        tok_list = tokens.tolist()
+        for s, t in zip(seqs, tok_list):
+            s.tokens.append(int(t))
+            if params.stop_token_id is not None and int(t) == params.stop_token_id:
+                s.finished = True
+
+    # -------------------------------------------------------------- generate
+    @torch.no_grad()
+    def generate(self, prompts: list[list[int]], params: SamplingParams,
+                 pad_token_id: int = 0, merge_lora: bool = True) -> torch.Tensor:
+        """Sample params.n continuations per prompt.
+
+        Returns LongTensor [len(prompts)*n, max_tokens]: responses
+        right-padded with pad_token_id, matching the reference's
+        vllm_generate output contract (grpo_trainer.py:152-164).
+        Row order: prompt-major (prompt0 sample0..n-1, prompt1 ...).
+        """
+        was_training = self.model.training
+        self.model.eval()
+        if merge_lora:
+            merge_for_rollout(self.model)
+        try:
+            waiting: list[SeqState] = []
+            uid = 0
+            for pi, p in enumerate(prompts):
+                for j in range(params.n):
+                    waiting.append(SeqState(uid, p, out_index=pi * params.n + j))
+                    uid += 1
+            done: list[SeqState] = []
+            active: list[SeqState] = []
+            self._sample_step = 0
+            steps = 0
+            max_steps_guard = params.max_tokens + 8
+
+            def admit():
+                """Move waiting → active, allocating each sequence's full page
+                budget (prompt + max_tokens) up front: no oversubscription, so
+                decode can never dead-lock on pages mid-flight."""
+                admitted = []
+                while waiting and len(active) + len(admitted) < self.max_num_seqs:
+                    s = waiting[-1]
+                    ps = self.pool.page_size
+                    need = (len(s) + params.max_tokens + ps - 1) // ps
+                    if need > self.pool.free_pages:
+                        break
+                    waiting.pop()
+                    s.ensure_capacity(self.pool, len(s) + params.max_tokens)
+                    admitted.append(s)
+                return admitted
+
+            while waiting or active:
+                fresh = admit()
+                if fresh:
+                    self._prefill(fresh, params)
+                    active.extend(fresh)
+                # retire finished / length-capped sequences, free their pages
+                still = []
+                for s in active:
+                    if s.finished or len(s.response) >= params.max_tokens:
+                        self.pool.free(s.pages)
+                        s.pages = []
+                        done.append(s)
+                    else:
+                        still.append(s)
+                active = still
+                if active:
+                    self._decode_step(active, params)
+                steps += 1
+                if steps > (max_steps_guard + len(prompts) * params.n):
+                    raise RuntimeError("sampler scheduling did not converge")
+
+            out = torch.full((len(prompts) * params.n, params.max_tokens), pad_token_id,
+                             dtype=torch.long)
+            for s in done:
+                resp = s.response[: params.max_tokens]
+                if resp:
+                    out[s.out_index, : len(resp)] = torch.tensor(resp, dtype=torch.long)
+            return out
+        finally:
+            if merge_lora:
+                unmerge(self.model)
+            if was_training:
+                self.model.train()

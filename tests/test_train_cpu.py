@@ -1,0 +1,127 @@
+"""End-to-end CPU training tests — one per algorithm, tiny model.
+
+test_reinforce_tiny_constant_reward is BASELINE.json config #1 (the
+permanent plumbing integration test: 2-layer 128-dim model, CPU eager
+generate, constant reward)."""
+import os
+
+import pytest
+import torch
+
+from nanorlhf_amd.algos import (GRPOConfig, PPOConfig, RAFTConfig, ReinforceConfig,
+                                RemaxConfig, RLOOConfig, finetune_value_model)
+from nanorlhf_amd.algos import grpo, ppo, raft, reinforce, remax, rloo
+from nanorlhf_amd.data import hh_shaped_prompts
+from nanorlhf_amd.models import CausalLM, ScalarHeadModel
+from nanorlhf_amd.rewards import constant_reward
+
+
+def _mk(cfg_cls, tmp_path, **kw):
+    base = dict(model_preset="tiny", dtype="float32", use_lora=True, lora_r=4,
+                lora_alpha=8, per_device_train_batch_size=2,
+                gradient_accumulation_steps=2, num_mini_batches=2,
+                total_episodes=16, response_length=6, temperature=1.0,
+                stop_token_id=1, output_dir=str(tmp_path),
+                gradient_checkpointing=False, score_token_budget=512,
+                missing_eos_penalty=1.0, save_steps=0)
+    base.update(kw)
+    return cfg_cls(**base)
+
+
+def _models(seed=0):
+    torch.manual_seed(seed)
+    policy = CausalLM.from_preset("tiny")
+    ref = CausalLM.from_preset("tiny")
+    ref.load_state_dict(policy.state_dict())
+    return policy, ref
+
+
+def _varied_reward(seqs):
+    # content-dependent reward so within-group advantages are non-degenerate
+    return torch.tensor([(sum(s) + len(s)) % 5 - 2.0 for s in seqs])
+
+
+def test_reinforce_tiny_constant_reward(tmp_path):
+    cfg = _mk(ReinforceConfig, tmp_path, save_steps=1)
+    policy, ref = _models()
+    prompts = hh_shaped_prompts(32, 1024, min_len=4, max_len=12)
+    tr = reinforce.make_trainer(cfg, policy, ref, lambda s: constant_reward(s), prompts)
+    tr.train(num_updates=2)
+    assert tr.episode == 16
+    assert os.path.isdir(os.path.join(tmp_path, "checkpoint-2"))
+    assert os.path.exists(os.path.join(tmp_path, "checkpoint-2", "trainer_state.json"))
+    assert os.path.exists(os.path.join(tmp_path, "metrics.jsonl"))
+
+
+def test_grpo_tiny(tmp_path):
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2)
+    policy, ref = _models(1)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=1)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    before = {n: p.clone() for n, p in tr.policy.named_parameters() if p.requires_grad}
+    tr.train(num_updates=1)
+    changed = any(not torch.equal(before[n], p)
+                  for n, p in tr.policy.named_parameters() if n in before)
+    assert changed, "update must move trainable params"
+
+
+def test_grpo_sparse_filter(tmp_path):
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2, sparse_filter=True,
+              train_token_budget=64)
+    policy, ref = _models(2)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=2)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_rloo_tiny(tmp_path):
+    cfg = _mk(RLOOConfig, tmp_path, sample_n=2)
+    policy, ref = _models(3)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=3)
+    tr = rloo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_remax_tiny(tmp_path):
+    cfg = _mk(RemaxConfig, tmp_path)
+    policy, ref = _models(4)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=4)
+    tr = remax.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_raft_tiny(tmp_path):
+    cfg = _mk(RAFTConfig, tmp_path, sample_n=2)
+    policy, ref = _models(5)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=5)
+    tr = raft.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_ppo_tiny_with_value_init(tmp_path):
+    cfg = _mk(PPOConfig, tmp_path, save_steps=1)
+    policy, ref = _models(6)
+    torch.manual_seed(6)
+    vm = ScalarHeadModel.from_preset("tiny", num_labels=1, bidirectional=False)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=6)
+    tr = ppo.make_trainer(cfg, policy, ref, _varied_reward, prompts, value_model=vm)
+    stats = finetune_value_model(tr, num_prompts=8, epochs=2, minibatch_rows=4)
+    assert stats["epochs_ran"] >= 1
+    tr.train(num_updates=1)
+    # PPO checkpoints include value_model/ (ppo_trainer.py:413-416)
+    assert os.path.isdir(os.path.join(tmp_path, "checkpoint-1", "value_model"))
+
+
+def test_model_reward_path(tmp_path):
+    from nanorlhf_amd.rewards import ModelReward
+    torch.manual_seed(7)
+    rm = ScalarHeadModel.from_preset("rm-tiny")
+    reward = ModelReward(rm, device="cpu", token_budget=128)
+    seqs = [list(range(2, 12)), list(range(2, 30))]
+    s = reward(seqs)
+    assert s.shape == (2,)
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2)
+    policy, ref = _models(7)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=7)
+    tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
+    tr.train(num_updates=1)
