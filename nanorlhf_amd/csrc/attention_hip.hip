@@ -1,0 +1,472 @@
+#include "hip/hip_runtime.h"
+// Varlen causal flash attention, forward + backward, MFMA bf16 for gfx950.
+//
+// Forward (fa_fwd_varlen): one 256-thread workgroup (4 waves) per
+// (sequence, q-head, 64-row q-tile); each wave owns 16 q rows.  KV walks in
+// 32-token tiles staged cooperatively in LDS (K row-major for the QK^T
+// B-fragment, V transposed so the PV B-fragment is a contiguous
+// ds_read_b128).  QK^T and PV are v_mfma_f32_16x16x32_bf16; softmax is
+// online in fp32 with per-row running (m, s); P round-trips through a
+// per-wave LDS buffer to convert the C-layout into an A-fragment.
+// Returns (o, lse[T, Hq] fp32) — lse feeds the backward's P recompute.
+//
+// Backward (fa_bwd_varlen): FA2 structure — one workgroup per (sequence,
+// kv-head, 32-token kv-tile); the G grouped q-heads x overlapping q-tiles
+// are strided across the 4 waves; dK/dV accumulate in AGPRs across the pair
+// loop and leave via fp32 atomics (workgroup-exclusive per kv tile, so only
+// the 4 waves contend); dQ contributions leave via fp32 atomics as in FA2.
+// dV/dK use v_mfma_f32_32x32x16_bf16 with P^T / dS^T staged per-wave.
+//
+// Replaces the reference's flash_attention_2 dependency
+// (GRPO/grpo.py:219,223); packed varlen replaces its pad-mask forward
+// (grpo_trainer.py:90-120).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16_t __attribute__((ext_vector_type(16)));
+
+DEVINL f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+DEVINL f32x16_t mfma32x32x16(bf16x8 a, bf16x8 b, f32x16_t c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+DEVINL void lds_fence_wave() { asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory"); }
+
+DEVINL __bf16 f2bf16t(float f) {
+  union { short s; __bf16 b; } u;
+  u.s = f2bf(f);
+  return u.b;
+}
+
+// =========================================================== FORWARD =====
+// q [T,Hq,D] bf16, k/v [T,Hkv,D] bf16, cu [B+1] i32 -> o [T,Hq,D], lse [T,Hq] f32
+template <int D, bool CAUSAL>
+__global__ void fa_fwd_kernel(const short* __restrict__ q,
+                              const short* __restrict__ k,
+                              const short* __restrict__ v,
+                              const int* __restrict__ cu,
+                              short* __restrict__ o,
+                              float* __restrict__ lse,
+                              int Hq, int Hkv, float scale) {
+  constexpr int QTILE = 64;   // per workgroup; 16 per wave
+  constexpr int KTILE = 32;
+  constexpr int NC = D / 32;  // K-chunks per mfma row
+  const int seq = blockIdx.y;
+  const int h = blockIdx.z;
+  const int kvh = h / (Hq / Hkv);
+  const int s0 = cu[seq], s1 = cu[seq + 1];
+  const int len = s1 - s0;
+  const int q0 = blockIdx.x * QTILE;
+  if (q0 >= len) return;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int hi = lane >> 4;           // 0..3
+  const int lo = lane & 15;
+  const int qr0 = q0 + wid * 16;      // this wave's first q row (tile-local)
+
+  __shared__ short Kt[KTILE * D];       // row-major [tok][d]
+  __shared__ short Vt[D * KTILE];       // transposed [d][tok]
+  __shared__ short Pb[4][16 * KTILE];   // per-wave P [q][kv]
+
+  // ---- Q fragments: lane holds Q[lo][32c + 8hi + j], scaled later -------
+  bf16x8 qf[NC];
+  const bool wave_active = qr0 < len;
+  {
+    const int qrow = qr0 + lo;
+    const long base = ((long)(s0 + min(qrow, len - 1)) * Hq + h) * D;
+#pragma unroll
+    for (int c = 0; c < NC; c++) {
+      s16x8 raw = *reinterpret_cast<const s16x8*>(q + base + 32 * c + 8 * hi);
+      qf[c] = *reinterpret_cast<bf16x8*>(&raw);
+    }
+  }
+
+  float m[4], s_[4];
+  f32x4 acc_o[D / 16];
+#pragma unroll
+  for (int r = 0; r < 4; r++) { m[r] = -INFINITY; s_[r] = 0.f; }
+#pragma unroll
+  for (int t = 0; t < D / 16; t++) acc_o[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = CAUSAL ? min(len, q0 + QTILE) : len;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KTILE) {
+    // ---- cooperative K/V stage ----------------------------------------
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < KTILE * D / 8; idx += 256) {
+      const int tok = idx / (D / 8);
+      const int d0 = (idx % (D / 8)) * 8;
+      const int kvi = kv0 + tok;
+      s16x8 kk{}, vv{};
+      if (kvi < len) {
+        const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
+        kk = *reinterpret_cast<const s16x8*>(k + b);
+        vv = *reinterpret_cast<const s16x8*>(v + b);
+      }
+      *reinterpret_cast<s16x8*>(&Kt[tok * D + d0]) = kk;
+#pragma unroll
+      for (int j = 0; j < 8; j++) Vt[(d0 + j) * KTILE + tok] = vv[j];
+    }
+    __syncthreads();
+    if (!wave_active) continue;
+
+    // ---- S = Q K^T : two 16-col subtiles -------------------------------
+    f32x4 sc[2];
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int c = 0; c < NC; c++) {
+        s16x8 raw = *reinterpret_cast<const s16x8*>(&Kt[(16 * n + lo) * D + 32 * c + 8 * hi]);
+        sc[n] = mfma16x16x32(qf[c], *reinterpret_cast<bf16x8*>(&raw), sc[n]);
+      }
+    }
+    // ---- mask + online softmax ----------------------------------------
+    // C layout: row (q) = 4*hi + reg, col (kv) = lo + 16*n
+    float pm[2][4];
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int qi = qr0 + 4 * hi + r;
+        const int ki = kv0 + 16 * n + lo;
+        const bool ok = (qi < len) && (ki < len) && (!CAUSAL || ki <= qi);
+        pm[n][r] = ok ? sc[n][r] * scale : -INFINITY;
+      }
+    }
+    float rowmax[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float x = fmaxf(pm[0][r], pm[1][r]);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off));
+      rowmax[r] = x;
+    }
+    float fac[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const float mn = fmaxf(m[r], rowmax[r]);
+      fac[r] = (m[r] == -INFINITY) ? 0.f : __expf(m[r] - mn);
+      m[r] = mn;
+    }
+    // P = exp(score - m), row sums, stage P to per-wave LDS
+    float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const float p = (pm[n][r] == -INFINITY) ? 0.f : __expf(pm[n][r] - m[r]);
+        rowsum[r] += p;
+        Pb[wid][(4 * hi + r) * KTILE + 16 * n + lo] = f2bf(p);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float x = rowsum[r];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) x += __shfl_xor(x, off);
+      s_[r] = s_[r] * fac[r] + x;
+    }
+    // rescale O accumulators (row = 4*hi + reg)
+#pragma unroll
+    for (int t = 0; t < D / 16; t++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) acc_o[t][r] *= fac[r];
+    }
+    lds_fence_wave();
+    // ---- PV: A = P (LDS), B = V^T slice (contiguous via Vt) ------------
+    bf16x8 pf;
+    {
+      s16x8 raw = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 8 * hi]);
+      pf = *reinterpret_cast<bf16x8*>(&raw);
+    }
+#pragma unroll
+    for (int t = 0; t < D / 16; t++) {
+      s16x8 raw = *reinterpret_cast<const s16x8*>(&Vt[(t * 16 + lo) * KTILE + 8 * hi]);
+      acc_o[t] = mfma16x16x32(pf, *reinterpret_cast<bf16x8*>(&raw), acc_o[t]);
+    }
+  }
+
+  if (!wave_active) return;
+  // ---- epilogue: O /= s, store o + lse --------------------------------
+#pragma unroll
+  for (int r = 0; r < 4; r++) {
+    const int qi = qr0 + 4 * hi + r;
+    if (qi >= len) continue;
+    const float inv = s_[r] > 0.f ? 1.f / s_[r] : 0.f;
+    const long base = ((long)(s0 + qi) * Hq + h) * D;
+#pragma unroll
+    for (int t = 0; t < D / 16; t++)
+      o[base + t * 16 + lo] = f2bf(acc_o[t][r] * inv);
+    if (lo == 0)
+      lse[(long)(s0 + qi) * Hq + h] = m[r] + __logf(fmaxf(s_[r], 1e-30f));
+  }
+}
+
+// ====================================================== BWD: D = rowsum ==
+__global__ void fa_bwd_preprocess_kernel(const short* __restrict__ dout,
+                                         const short* __restrict__ o,
+                                         float* __restrict__ drow,
+                                         int Hq, int D) {
+  const long row = blockIdx.x;  // t * Hq + h
+  const int lane = threadIdx.x;
+  const long base = row * D;
+  float acc = 0.f;
+  for (int d = lane; d < D; d += 64)
+    acc += bf2f(dout[base + d]) * bf2f(o[base + d]);
+  acc = wave_sum(acc);
+  if (lane == 0) drow[row] = acc;
+}
+
+// =========================================================== BACKWARD ====
+template <int D, bool CAUSAL>
+__global__ void fa_bwd_kernel(const short* __restrict__ dout,
+                              const short* __restrict__ q,
+                              const short* __restrict__ k,
+                              const short* __restrict__ v,
+                              const float* __restrict__ lse,
+                              const float* __restrict__ drow,
+                              const int* __restrict__ cu,
+                              float* __restrict__ dqf,   // [T,Hq,D] f32
+                              float* __restrict__ dkf,   // [T,Hkv,D] f32
+                              float* __restrict__ dvf,
+                              int Hq, int Hkv, float scale) {
+  constexpr int KTILE = 32;
+  constexpr int NC = D / 32;
+  const int seq = blockIdx.y;
+  const int kvh = blockIdx.z;
+  const int G = Hq / Hkv;
+  const int s0 = cu[seq], s1 = cu[seq + 1];
+  const int len = s1 - s0;
+  const int kv0 = blockIdx.x * KTILE;
+  if (kv0 >= len) return;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int hi = lane >> 4, lo = lane & 15;
+  const int hi5 = lane >> 5, lo5 = lane & 31;
+
+  __shared__ short Kt[KTILE * D];     // [tok][d]
+  __shared__ short KT[D * KTILE];     // [d][tok]
+  __shared__ short Vt[KTILE * D];     // [tok][d]
+  __shared__ short QT[4][D * 16];     // per-wave Q^T [d][q]
+  __shared__ short dOT[4][D * 16];    // per-wave dO^T [d][q]
+  __shared__ short PT[4][KTILE * 16];   // per-wave P^T [kv][q]
+  __shared__ short dSb[4][16 * KTILE];  // per-wave dS [q][kv]
+  __shared__ short dST[4][KTILE * 16];  // per-wave dS^T [kv][q]
+
+  // cooperative stage of K (both layouts) and V
+  for (int idx = threadIdx.x; idx < KTILE * D / 8; idx += 256) {
+    const int tok = idx / (D / 8);
+    const int d0 = (idx % (D / 8)) * 8;
+    const int kvi = kv0 + tok;
+    s16x8 kk{}, vv{};
+    if (kvi < len) {
+      const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
+      kk = *reinterpret_cast<const s16x8*>(k + b);
+      vv = *reinterpret_cast<const s16x8*>(v + b);
+    }
+    *reinterpret_cast<s16x8*>(&Kt[tok * D + d0]) = kk;
+    *reinterpret_cast<s16x8*>(&Vt[tok * D + d0]) = vv;
+#pragma unroll
+    for (int j = 0; j < 8; j++) KT[(d0 + j) * KTILE + tok] = kk[j];
+  }
+  __syncthreads();
+
+  // dV/dK accumulators: 32x32 C tiles over D/32 column chunks
+  f32x16_t acc_dv[NC], acc_dk[NC];
+#pragma unroll
+  for (int c = 0; c < NC; c++) {
+    acc_dv[c] = f32x16_t{};
+    acc_dk[c] = f32x16_t{};
+  }
+
+  const int qt0 = CAUSAL ? (kv0 / 16) : 0;
+  const int nqt = (len + 15) / 16;
+  const int npairs = G * (nqt - qt0);
+
+  for (int p = wid; p < npairs; p += 4) {
+    const int g = p % G;
+    const int qt = qt0 + p / G;
+    const int h = kvh * G + g;
+    const int r0 = qt * 16;
+
+    // ---- load Q / dO fragments + stage transposed copies --------------
+    bf16x8 qfr[NC], dofr[NC];
+    {
+      const int qrow = r0 + lo;
+      const long base = ((long)(s0 + min(qrow, len - 1)) * Hq + h) * D;
+      const bool ok = qrow < len;
+#pragma unroll
+      for (int c = 0; c < NC; c++) {
+        s16x8 rq{}, rd{};
+        if (ok) {
+          rq = *reinterpret_cast<const s16x8*>(q + base + 32 * c + 8 * hi);
+          rd = *reinterpret_cast<const s16x8*>(dout + base + 32 * c + 8 * hi);
+        }
+        qfr[c] = *reinterpret_cast<bf16x8*>(&rq);
+        dofr[c] = *reinterpret_cast<bf16x8*>(&rd);
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          QT[wid][(32 * c + 8 * hi + j) * 16 + lo] = rq[j];
+          dOT[wid][(32 * c + 8 * hi + j) * 16 + lo] = rd[j];
+        }
+      }
+    }
+
+    // ---- S and dP (both 16x32, C layout row=q col=kv) ------------------
+    f32x4 sc[2], dp[2];
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+      dp[n] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int c = 0; c < NC; c++) {
+        s16x8 rk = *reinterpret_cast<const s16x8*>(&Kt[(16 * n + lo) * D + 32 * c + 8 * hi]);
+        s16x8 rv = *reinterpret_cast<const s16x8*>(&Vt[(16 * n + lo) * D + 32 * c + 8 * hi]);
+        sc[n] = mfma16x16x32(qfr[c], *reinterpret_cast<bf16x8*>(&rk), sc[n]);
+        dp[n] = mfma16x16x32(dofr[c], *reinterpret_cast<bf16x8*>(&rv), dp[n]);
+      }
+    }
+
+    // ---- P, dS (fp32, C layout) + stage P^T / dS / dS^T ----------------
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int qi = r0 + 4 * hi + r;
+        const int ki = kv0 + 16 * n + lo;
+        const bool ok = (qi < len) && (ki < len) && (!CAUSAL || ki <= qi);
+        float pv = 0.f, dsv = 0.f;
+        if (ok) {
+          const long li = (long)(s0 + qi) * Hq + h;
+          pv = __expf(sc[n][r] * scale - lse[li]);
+          dsv = pv * (dp[n][r] - drow[li]) * scale;
+        }
+        PT[wid][(16 * n + lo) * 16 + 4 * hi + r] = f2bf(pv);
+        dSb[wid][(4 * hi + r) * KTILE + 16 * n + lo] = f2bf(dsv);
+        dST[wid][(16 * n + lo) * 16 + 4 * hi + r] = f2bf(dsv);
+      }
+    }
+    lds_fence_wave();
+
+    // ---- dV += P^T dO ; dK += dS^T Q (32x32x16, K = 16 q rows) ---------
+    {
+      s16x8 rp = *reinterpret_cast<const s16x8*>(&PT[wid][lo5 * 16 + 8 * hi5]);
+      s16x8 rs = *reinterpret_cast<const s16x8*>(&dST[wid][lo5 * 16 + 8 * hi5]);
+      bf16x8 pA = *reinterpret_cast<bf16x8*>(&rp);
+      bf16x8 sA = *reinterpret_cast<bf16x8*>(&rs);
+#pragma unroll
+      for (int c = 0; c < NC; c++) {
+        s16x8 rdo = *reinterpret_cast<const s16x8*>(&dOT[wid][(32 * c + lo5) * 16 + 8 * hi5]);
+        s16x8 rqt = *reinterpret_cast<const s16x8*>(&QT[wid][(32 * c + lo5) * 16 + 8 * hi5]);
+        acc_dv[c] = mfma32x32x16(pA, *reinterpret_cast<bf16x8*>(&rdo), acc_dv[c]);
+        acc_dk[c] = mfma32x32x16(sA, *reinterpret_cast<bf16x8*>(&rqt), acc_dk[c]);
+      }
+    }
+
+    // ---- dQ = dS K (16x16x32 over d tiles) -----------------------------
+    {
+      s16x8 rds = *reinterpret_cast<const s16x8*>(&dSb[wid][lo * KTILE + 8 * hi]);
+      bf16x8 dsA = *reinterpret_cast<bf16x8*>(&rds);
+#pragma unroll
+      for (int t = 0; t < D / 16; t++) {
+        s16x8 rkt = *reinterpret_cast<const s16x8*>(&KT[(t * 16 + lo) * KTILE + 8 * hi]);
+        f32x4 dq = mfma16x16x32(dsA, *reinterpret_cast<bf16x8*>(&rkt), f32x4{0.f, 0.f, 0.f, 0.f});
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const int qi = r0 + 4 * hi + r;
+          if (qi < len)
+            atomicAdd(&dqf[((long)(s0 + qi) * Hq + h) * D + t * 16 + lo], dq[r]);
+        }
+      }
+    }
+  }
+
+  // ---- drain dV/dK (32x32 C layout: row=kv, col=d) ----------------------
+#pragma unroll
+  for (int c = 0; c < NC; c++) {
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int kvi = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+      const int d = 32 * c + lo5;
+      if (kvi < len) {
+        const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d;
+        atomicAdd(&dvf[b], acc_dv[c][r]);
+        atomicAdd(&dkf[b], acc_dk[c][r]);
+      }
+    }
+  }
+}
+
+// ======================================================== host wrappers ==
+std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, torch::Tensor cu_seqlens,
+                                         long max_seqlen, double scale, bool causal) {
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt);
+  const long T = q.size(0);
+  const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
+  const int B = cu_seqlens.size(0) - 1;
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
+  if (T == 0) return {o, lse};
+  const int qtiles = (int)((max_seqlen + 63) / 64);
+  dim3 grid(qtiles, B, Hq), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+#define FWD_LAUNCH(DD, CC)                                                     \
+  hipLaunchKernelGGL((fa_fwd_kernel<DD, CC>), grid, block, 0, stream,          \
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),   \
+                     (const short*)v.data_ptr(), cu_seqlens.data_ptr<int>(),   \
+                     (short*)o.data_ptr(), lse.data_ptr<float>(), Hq, Hkv,     \
+                     (float)scale)
+  if (D == 128) { if (causal) FWD_LAUNCH(128, true); else FWD_LAUNCH(128, false); }
+  else if (D == 64) { if (causal) FWD_LAUNCH(64, true); else FWD_LAUNCH(64, false); }
+  else if (D == 32) { if (causal) FWD_LAUNCH(32, true); else FWD_LAUNCH(32, false); }
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
+#undef FWD_LAUNCH
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> fa_bwd_varlen(torch::Tensor dout, torch::Tensor q,
+                                         torch::Tensor k, torch::Tensor v,
+                                         torch::Tensor o, torch::Tensor lse,
+                                         torch::Tensor cu_seqlens, long max_seqlen,
+                                         double scale, bool causal) {
+  TORCH_CHECK(dout.scalar_type() == torch::kBFloat16 && dout.is_contiguous());
+  const long T = q.size(0);
+  const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
+  const int B = cu_seqlens.size(0) - 1;
+  auto opts = q.options().dtype(torch::kFloat32);
+  auto dqf = torch::zeros({T, (long)Hq, (long)D}, opts);
+  auto dkf = torch::zeros({T, (long)Hkv, (long)D}, opts);
+  auto dvf = torch::zeros({T, (long)Hkv, (long)D}, opts);
+  if (T == 0) return {dqf.to(torch::kBFloat16), dkf.to(torch::kBFloat16),
+                      dvf.to(torch::kBFloat16)};
+  auto drow = torch::empty({T, (long)Hq}, opts);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fa_bwd_preprocess_kernel, dim3(T * Hq), dim3(64), 0, stream,
+                     (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
+                     drow.data_ptr<float>(), Hq, D);
+  const int kvtiles = (int)((max_seqlen + 31) / 32);
+  dim3 grid(kvtiles, B, Hkv), block(256);
+#define BWD_LAUNCH(DD, CC)                                                     \
+  hipLaunchKernelGGL((fa_bwd_kernel<DD, CC>), grid, block, 0, stream,          \
+                     (const short*)dout.data_ptr(), (const short*)q.data_ptr(),\
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),   \
+                     lse.data_ptr<float>(), drow.data_ptr<float>(),            \
+                     cu_seqlens.data_ptr<int>(), dqf.data_ptr<float>(),        \
+                     dkf.data_ptr<float>(), dvf.data_ptr<float>(), Hq, Hkv,    \
+                     (float)scale)
+  if (D == 128) { if (causal) BWD_LAUNCH(128, true); else BWD_LAUNCH(128, false); }
+  else if (D == 64) { if (causal) BWD_LAUNCH(64, true); else BWD_LAUNCH(64, false); }
+  else if (D == 32) { if (causal) BWD_LAUNCH(32, true); else BWD_LAUNCH(32, false); }
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
+#undef BWD_LAUNCH
+  HIP_CHECK_LAST();
+  return {dqf.to(torch::kBFloat16), dkf.to(torch::kBFloat16), dvf.to(torch::kBFloat16)};
+}

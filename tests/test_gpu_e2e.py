@@ -1,0 +1,75 @@
+"""GPU end-to-end: sampler engine on the HIP path vs full-forward oracle,
+and a short GRPO training run on a small bf16 model."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def _small_model(seed=0, layers=2, vocab=4096):
+    from nanorlhf_amd.models import CausalLM, get_config
+    torch.manual_seed(seed)
+    cfg = get_config("qwen2.5-1.5b", num_layers=layers, vocab_size=vocab)
+    return CausalLM(cfg).to(DEV).to(torch.bfloat16)
+
+
+def test_sampler_gpu_greedy_matches_oracle():
+    from nanorlhf_amd.models import CausalLM, pack_sequences
+    from nanorlhf_amd.sampler import SamplerEngine, SamplingParams
+    m = _small_model().eval()
+    eng = SamplerEngine(m, kv_pool_tokens=16384, page_size=16)
+    prompts = [torch.randint(2, 4096, (n,)).tolist() for n in (5, 21, 40)]
+    params = SamplingParams(n=1, temperature=0.0, top_p=1.0, max_tokens=12, seed=1)
+    out = eng.generate(prompts, params)
+
+    def oracle(prompt, steps):
+        toks = list(prompt)
+        for _ in range(steps):
+            ids, cu, mx, pos = pack_sequences([torch.tensor(toks)], device=DEV)
+            h = m(ids, CausalLM.train_ctx(cu, mx, pos))
+            toks.append(int(m.logits(h[-1:]).argmax()))
+        return toks[len(prompt):]
+
+    for i, p in enumerate(prompts):
+        want = oracle(p, 12)
+        got = out[i].tolist()
+        # bf16 paged decode vs recompute may diverge after many steps if two
+        # logits are near-ties; require the first 8 to match exactly
+        assert got[:8] == want[:8], (i, got, want)
+
+
+def test_grpo_gpu_short_training():
+    from nanorlhf_amd.algos import grpo
+    from nanorlhf_amd.algos.grpo import GRPOConfig
+    from nanorlhf_amd.data import hh_shaped_prompts
+    from nanorlhf_amd.models import CausalLM, get_config
+
+    torch.manual_seed(0)
+    cfg_m = get_config("qwen2.5-1.5b", num_layers=2, vocab_size=4096)
+    policy = CausalLM(cfg_m)
+    ref = CausalLM(cfg_m)
+    ref.load_state_dict(policy.state_dict())
+    cfg = GRPOConfig(model_preset="custom", dtype="bfloat16", use_lora=True,
+                     lora_r=8, lora_alpha=16, per_device_train_batch_size=2,
+                     gradient_accumulation_steps=2, num_mini_batches=2,
+                     total_episodes=16, sample_n=2, response_length=16,
+                     temperature=1.0, stop_token_id=1,
+                     output_dir="/tmp/nanorlhf_gpu_grpo",
+                     gradient_checkpointing=True, score_token_budget=4096,
+                     missing_eos_penalty=1.0)
+    prompts = hh_shaped_prompts(16, 4096, min_len=8, max_len=24)
+
+    def reward(seqs):
+        return torch.tensor([(sum(s) + len(s)) % 5 - 2.0 for s in seqs])
+
+    tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
+    before = {n: p.clone() for n, p in tr.policy.named_parameters() if p.requires_grad}
+    tr.train(num_updates=1)
+    moved = any(not torch.equal(before[n], p)
+                for n, p in tr.policy.named_parameters() if n in before)
+    assert moved
+    m = tr._last_metrics
+    assert all(torch.isfinite(torch.tensor(float(v))) for k, v in m.items()
+               if isinstance(v, (int, float))), m
