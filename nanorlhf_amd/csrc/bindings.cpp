@@ -5,7 +5,7 @@
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor invrms);
-void rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions, double sign);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions, double sign);
 torch::Tensor swiglu_fwd(torch::Tensor gu);
 torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu);
 // logprob.hip

@@ -93,7 +93,8 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
 // ----------------------------------------------------------------- RoPE
 // x [T, Hh, D] bf16 in-place; table [maxpos, D] f32 (cos | sin halves);
 // positions [T] i64.  sign=-1 applies the inverse rotation (backward).
-__global__ void rope_kernel(short* __restrict__ x,
+__global__ void rope_kernel(const short* __restrict__ x,
+                            short* __restrict__ y,
                             const float* __restrict__ table,
                             const long* __restrict__ pos,
                             int T, int Hh, int D, float sign) {
@@ -105,7 +106,8 @@ __global__ void rope_kernel(short* __restrict__ x,
   const int g = idx % pairs4;
   const int h = (idx / pairs4) % Hh;
   const long t = idx / ((long)pairs4 * Hh);
-  short* base = x + (t * Hh + h) * (long)D;
+  const short* base = x + (t * Hh + h) * (long)D;
+  short* obase = y + (t * Hh + h) * (long)D;
   const float* tb = table + pos[t] * (long)D;
   s16x4 x1 = *reinterpret_cast<const s16x4*>(base + g * 4);
   s16x4 x2 = *reinterpret_cast<const s16x4*>(base + D / 2 + g * 4);
@@ -119,8 +121,8 @@ __global__ void rope_kernel(short* __restrict__ x,
     o1[j] = f2bf(a * cs - b * sn);
     o2[j] = f2bf(b * cs + a * sn);
   }
-  *reinterpret_cast<s16x4*>(base + g * 4) = o1;
-  *reinterpret_cast<s16x4*>(base + D / 2 + g * 4) = o2;
+  *reinterpret_cast<s16x4*>(obase + g * 4) = o1;
+  *reinterpret_cast<s16x4*>(obase + D / 2 + g * 4) = o2;
 }
 
 // ----------------------------------------------------------------- SwiGLU
@@ -211,8 +213,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dwf.to(torch::kBFloat16)};
 }
 
-void rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions,
-              double sign) {
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions,
+                       double sign) {
   check_bf16(x, "x");
   TORCH_CHECK(table.scalar_type() == torch::kFloat32, "rope table must be fp32");
   TORCH_CHECK(positions.scalar_type() == torch::kLong, "positions must be int64");
@@ -223,11 +225,14 @@ void rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions,
   const long total = (long)T * Hh * (D / 8);
   const int block = 256;
   const long grid = (total + block - 1) / block;
+  auto y = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(block), 0, stream,
-                     (short*)x.data_ptr(), table.data_ptr<float>(),
+                     (const short*)x.data_ptr(), (short*)y.data_ptr(),
+                     table.data_ptr<float>(),
                      positions.data_ptr<long>(), T, Hh, D, (float)sign);
   HIP_CHECK_LAST();
+  return y;
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor gu) {

@@ -77,7 +77,7 @@ __global__ void ce_rowstats_kernel(const short* __restrict__ logits,
 }
 
 // dlogits (in place over logits, bf16):
-//   d_j = g * (softmax_j - onehot_j) * inv_temp
+//   d_j = g * (onehot_j - softmax_j) * inv_temp   (d logprob/d logit)
 __global__ void ce_backward_kernel(short* __restrict__ logits,
                                    const long* __restrict__ labels,
                                    const float* __restrict__ lse,
@@ -98,7 +98,7 @@ __global__ void ce_backward_kernel(short* __restrict__ logits,
   for (int j = 0; j < 8; j++) {
     float l = bf2f(v[j]) * inv_temp;
     float p = __expf(l - L);
-    float d = gr * (p - ((col0 + j) == lab ? 1.f : 0.f)) * inv_temp;
+    float d = gr * (((col0 + j) == lab ? 1.f : 0.f) - p) * inv_temp;
     o[j] = f2bf(d);
   }
   *reinterpret_cast<s16x8*>(lr + col0) = o;
@@ -117,7 +117,7 @@ __global__ void ce_backward_tail_kernel(short* __restrict__ logits,
   short* lr = logits + row * (long)V;
   float l = bf2f(lr[col]) * inv_temp;
   float p = __expf(l - lse[row]);
-  lr[col] = f2bf(g[row] * (p - (col == labels[row] ? 1.f : 0.f)) * inv_temp);
+  lr[col] = f2bf(g[row] * ((col == labels[row] ? 1.f : 0.f) - p) * inv_temp);
 }
 
 // ======================================================== host wrappers ==

@@ -97,13 +97,17 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   const int bstar = bin_star_sh;
   const float bin_lo = (float)bstar * (URANGE / NBINS) - URANGE;
   const float bin_hi = bin_lo + (URANGE / NBINS);
-  // ---- refinement: sub-histogram inside bin* ----------------------------
+  // ---- refinement: sub-histogram inside bin* (membership decided by the
+  // SAME coarse binning as pass 2, so boundary values like u == 0 land in
+  // the clamped top bin consistently) -----------------------------------
   for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist[i] = 0.f;
   __syncthreads();
   const float sub_scale = NBINS / (bin_hi - bin_lo);
   for (int i = threadIdx.x; i < V; i += SBLOCK) {
     float u = (bf2f(lr[i]) - m) * invT;
-    if (u >= bin_lo && u < bin_hi) {
+    int cb = (int)((u + URANGE) * (NBINS / URANGE));
+    cb = max(0, min(NBINS - 1, cb));
+    if (cb == bstar) {
       int b = (int)((u - bin_lo) * sub_scale);
       b = max(0, min(NBINS - 1, b));
       atomicAdd(&hist[b], __expf(u) * invS);
@@ -124,8 +128,12 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     u_thresh_sh = thresh;
   }
   __syncthreads();
-  const float u_thresh = u_thresh_sh;
-  const float kept_mass = mass_above_sh;
+  float u_thresh = u_thresh_sh;
+  float kept_mass = mass_above_sh;
+  if (kept_mass < 1e-9f) {  // degenerate: keep everything
+    u_thresh = -2.f * URANGE;
+    kept_mass = 1.f;
+  }
 
   // ---- pass 3: inverse-CDF over kept tokens (index order) ---------------
   const float r = hash_uniform(seed, step, (unsigned long long)row);

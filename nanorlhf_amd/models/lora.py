@@ -34,8 +34,9 @@ class LoRALinear(nn.Module):
         for p in self.base.parameters():
             p.requires_grad_(False)
         dtype = base.weight.dtype
-        self.lora_A = nn.Parameter(torch.zeros(r, base.in_features, dtype=dtype))
-        self.lora_B = nn.Parameter(torch.zeros(base.out_features, r, dtype=dtype))
+        dev = base.weight.device
+        self.lora_A = nn.Parameter(torch.zeros(r, base.in_features, dtype=dtype, device=dev))
+        self.lora_B = nn.Parameter(torch.zeros(base.out_features, r, dtype=dtype, device=dev))
         nn.init.kaiming_uniform_(self.lora_A, a=5 ** 0.5)
         self.scaling = alpha / r
         self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()

@@ -40,15 +40,12 @@ class _RopeFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, table, positions):
         ctx.save_for_backward(table, positions)
-        out = x.contiguous()
-        ext().rope_fwd(out, table, positions, 1.0)  # in place on the copy
-        return out
+        return ext().rope_fwd(x.contiguous(), table, positions, 1.0)
 
     @staticmethod
     def backward(ctx, dy):
         table, positions = ctx.saved_tensors
-        dx = dy.contiguous()
-        ext().rope_fwd(dx, table, positions, -1.0)
+        dx = ext().rope_fwd(dy.contiguous(), table, positions, -1.0)
         return dx, None, None
 
 
