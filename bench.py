@@ -110,8 +110,14 @@ def main():
         dist_.all_reduce(t, op=dist_.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    episodes_total = n_prompts * world * args.steps
+    local_batch = tr.sizes["local_batch_size"]  # actual prompts consumed/update
+    episodes_total = local_batch * world * args.steps
     eps_per_sec = episodes_total / elapsed
+    if rank == 0:
+        phases = tr.timers.snapshot_and_reset()
+        print(f"[bench] phase totals over {args.warmup}+{args.steps} updates: " +
+              ", ".join(f"{k}={v:.1f}s" for k, v in sorted(phases.items())
+                        if isinstance(v, float)), file=sys.stderr)
     ms_per_step = elapsed / args.steps * 1000.0
     if rank == 0:
         print(json.dumps({
@@ -130,7 +136,7 @@ def main():
             "config": {
                 "model": "qwen2.5-1.5b-instruct-arch (random init)",
                 "reward_model": "deberta-v3-large-shaped encoder (random init)",
-                "global_batch": n_prompts * world,
+                "global_batch": local_batch * world,
                 "sample_n": args.sample_n,
                 "seq_len": args.response_length,
                 "lora_r": 64,

@@ -41,8 +41,12 @@ class ModelReward:
                 chunk.append(i)
                 i += 1
             lens = [len(sequences[j]) for j in chunk]
-            ids = torch.tensor([t for j in chunk for t in sequences[j]], dtype=torch.long,
-                               device=self.device)
+            V = self.model.cfg.vocab_size
+            # the RM has its own vocab (the reference re-tokenizes detokenized
+            # strings with the RM tokenizer, grpo.py:180-187); for the raw-id
+            # path fold ids into the RM vocab deterministically
+            ids = torch.tensor([t % V for j in chunk for t in sequences[j]],
+                               dtype=torch.long, device=self.device)
             cu = torch.zeros(len(chunk) + 1, dtype=torch.int32, device=self.device)
             cu[1:] = torch.cumsum(torch.tensor(lens, dtype=torch.int32, device=self.device), 0)
             s = self.model.sequence_scores(ids, cu, max(lens))
