@@ -22,6 +22,7 @@ from .. import ops
 from ..models.qwen2 import AttnContext, CausalLM
 from ..models.lora import merge_for_rollout, unmerge
 from .cache import PagedKVCache, SeqState
+from .decode_batch import DecodeBatch
 
 
 @dataclass
@@ -36,7 +37,8 @@ class SamplingParams:
 
 class SamplerEngine:
     def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
-                 max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072):
+                 max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072,
+                 compact_interval: int = 16):
         self.model = model
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
@@ -44,6 +46,7 @@ class SamplerEngine:
                                             device=self.device, dtype=self.dtype)
         self.max_num_seqs = max_num_seqs
         self.prefill_chunk_tokens = prefill_chunk_tokens
+        self.compact_interval = compact_interval
 
     # ----------------------------------------------------------------- utils
     def _slots_for_range(self, seq: SeqState, start: int, end: int) -> list[int]:
@@ -90,33 +93,6 @@ class SamplerEngine:
                 if params.stop_token_id is not None and int(t) == params.stop_token_id:
                     s.finished = True
 
-    # --------------------------------------------------------------- decode
-    @torch.no_grad()
-    def _decode_step(self, seqs: list[SeqState], params: SamplingParams):
-        """One decode step over all active sequences."""
-        ps = self.pool.page_size
-        for s in seqs:
-            s.ensure_capacity(self.pool, len(s) + 1)
-        ids = torch.tensor([s.tokens[-1] for s in seqs], dtype=torch.long, device=self.device)
-        pos = torch.tensor([len(s) - 1 for s in seqs], dtype=torch.long, device=self.device)
-        slots = torch.tensor([s.slot_of(len(s) - 1, ps) for s in seqs], dtype=torch.long,
-                             device=self.device)
-        seq_lens = torch.tensor([len(s) for s in seqs], dtype=torch.int32, device=self.device)
-        max_pages = max(len(s.pages) for s in seqs)
-        bt = torch.zeros(len(seqs), max_pages, dtype=torch.int32)
-        for r, s in enumerate(seqs):
-            bt[r, : len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
-        bt = bt.to(self.device)
-        ctx = AttnContext(mode="decode", positions=pos, kv_caches=self.pool.layers,
-                          slots=slots, block_tables=bt, seq_lens=seq_lens)
-        hidden = self.model(ids, ctx)
-        tokens = self._sample_from_hidden(hidden, params)
-        tok_list = tokens.tolist()
-        for s, t in zip(seqs, tok_list):
-            s.tokens.append(int(t))
-            if params.stop_token_id is not None and int(t) == params.stop_token_id:
-                s.finished = True
-
     # -------------------------------------------------------------- generate
     @torch.no_grad()
     def generate(self, prompts: list[list[int]], params: SamplingParams,
@@ -140,17 +116,17 @@ class SamplerEngine:
                     waiting.append(SeqState(uid, p, out_index=pi * params.n + j))
                     uid += 1
             done: list[SeqState] = []
-            active: list[SeqState] = []
             self._sample_step = 0
-            steps = 0
-            max_steps_guard = params.max_tokens + 8
+            db = DecodeBatch(self.pool, [], params.max_tokens,
+                             params.stop_token_id, self.device)
+            outer_guard = 0
 
             def admit():
                 """Move waiting → active, allocating each sequence's full page
                 budget (prompt + max_tokens) up front: no oversubscription, so
                 decode can never dead-lock on pages mid-flight."""
                 admitted = []
-                while waiting and len(active) + len(admitted) < self.max_num_seqs:
+                while waiting and len(db) + len(admitted) < self.max_num_seqs:
                     s = waiting[-1]
                     ps = self.pool.page_size
                     need = (len(s) + params.max_tokens + ps - 1) // ps
@@ -161,25 +137,26 @@ class SamplerEngine:
                     admitted.append(s)
                 return admitted
 
-            while waiting or active:
+            while waiting or len(db):
                 fresh = admit()
                 if fresh:
                     self._prefill(fresh, params)
-                    active.extend(fresh)
-                # retire finished / length-capped sequences, free their pages
-                still = []
-                for s in active:
-                    if s.finished or len(s.response) >= params.max_tokens:
-                        self.pool.free(s.pages)
-                        s.pages = []
-                        done.append(s)
-                    else:
-                        still.append(s)
-                active = still
-                if active:
-                    self._decode_step(active, params)
-                steps += 1
-                if steps > (max_steps_guard + len(prompts) * params.n):
+                    db.extend(fresh)
+                # device-resident decode: no host sync inside the chunk
+                inner = 0
+                while len(db) and inner < self.compact_interval:
+                    ids, pos, slots, seq_lens, bt = db.step_inputs()
+                    ctx = AttnContext(mode="decode", positions=pos,
+                                      kv_caches=self.pool.layers, slots=slots,
+                                      block_tables=bt, seq_lens=seq_lens)
+                    hidden = self.model(ids, ctx)
+                    tokens = self._sample_from_hidden(hidden, params)
+                    db.commit(tokens, pad_token_id)
+                    inner += 1
+                done.extend(db.compact())
+                outer_guard += 1
+                if outer_guard > 8 * (params.max_tokens // self.compact_interval + 2) \
+                        + len(prompts) * params.n:
                     raise RuntimeError("sampler scheduling did not converge")
 
             out = torch.full((len(prompts) * params.n, params.max_tokens), pad_token_id,
