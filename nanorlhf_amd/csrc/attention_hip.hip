@@ -35,7 +35,7 @@ DEVINL f32x16_t mfma32x32x16(bf16x8 a, bf16x8 b, f32x16_t c) {
   return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
 }
 
-DEVINL void lds_fence_wave() { asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory"); }
+DEVINL void lds_fence_wave() { lds_fence_wave_kv(); }
 
 DEVINL __bf16 f2bf16t(float f) {
   union { short s; __bf16 b; } u;

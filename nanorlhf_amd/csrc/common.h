@@ -99,6 +99,10 @@ DEVINL float hash_uniform(unsigned long long seed, unsigned long long step,
   return (float)(h >> 40) * (1.0f / 16777216.0f);
 }
 
+// wave-private LDS write->read ordering (staging buffers written and read
+// by the same wave between barriers)
+DEVINL void lds_fence_wave_kv() { asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory"); }
+
 #define HIP_CHECK_LAST()                                                    \
   do {                                                                      \
     hipError_t e_ = hipGetLastError();                                      \

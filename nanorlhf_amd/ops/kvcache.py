@@ -20,14 +20,18 @@ from . import ext
 def kv_append(k: torch.Tensor, v: torch.Tensor, slots: torch.Tensor,
               k_cache: torch.Tensor, v_cache: torch.Tensor) -> None:
     """k/v: [T, Hkv, D] bf16; slots: [T] int64 flat slot index
-    (page * page_size + offset) into caches viewed as [num_pages*page_size, Hkv, D]."""
+    (page * page_size + offset).  K cache is [P, ps, Hkv, D]; V cache is
+    d-major per page, [P, Hkv, D, ps]."""
     if k.is_cuda:
         ext().kv_append(k.contiguous(), v.contiguous(), slots, k_cache, v_cache)
         return
+    ps = k_cache.shape[1]
     kc = k_cache.view(-1, *k_cache.shape[2:])
-    vc = v_cache.view(-1, *v_cache.shape[2:])
     kc[slots] = k
-    vc[slots] = v
+    pages = torch.div(slots, ps, rounding_mode="floor")
+    offs = slots % ps
+    # v_cache[page, h, d, off] = v[t, h, d]
+    v_cache[pages, :, :, offs] = v
 
 
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
@@ -47,7 +51,8 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Ten
     rep = Hq // Hkv
     out = torch.empty_like(q, dtype=torch.float32)
     kc = k_cache.view(-1, Hkv, D).float()
-    vc = v_cache.view(-1, Hkv, D).float()
+    # V d-major [P, Hkv, D, ps] -> flat [P*ps, Hkv, D]
+    vc = v_cache.permute(0, 3, 1, 2).reshape(-1, Hkv, D).float()
     for b in range(B):
         L = int(seq_lens[b])
         pages = block_tables[b, : (L + page - 1) // page].long()

@@ -18,10 +18,13 @@ class PagedKVCache:
         self.page_size = page_size
         self.num_pages = num_pages
         self.device = device
-        shape = (num_pages, page_size, cfg.num_kv_heads, cfg.head_dim)
+        kshape = (num_pages, page_size, cfg.num_kv_heads, cfg.head_dim)
+        # V is stored d-major within each page ([page][Hkv][D][page_size]) so
+        # the decode kernel's PV MFMA B-fragment is a contiguous 16-B load
+        vshape = (num_pages, cfg.num_kv_heads, cfg.head_dim, page_size)
         self.layers = [
-            (torch.zeros(shape, dtype=dtype, device=device),
-             torch.zeros(shape, dtype=dtype, device=device))
+            (torch.zeros(kshape, dtype=dtype, device=device),
+             torch.zeros(vshape, dtype=dtype, device=device))
             for _ in range(cfg.num_layers)
         ]
         self._free = list(range(num_pages - 1, -1, -1))
