@@ -158,7 +158,7 @@ def test_kv_append_and_paged_decode():
     cfg = get_config("qwen2.5-1.5b", num_layers=1)
     B, Hq, Hkv, D, ps = 5, cfg.num_heads, cfg.num_kv_heads, cfg.head_dim, 16
     kc = torch.zeros(64, ps, Hkv, D, dtype=torch.bfloat16, device=DEV)
-    vc = torch.zeros_like(kc)
+    vc = torch.zeros(64, Hkv, D, ps, dtype=torch.bfloat16, device=DEV)  # d-major pages
     lens = [7, 33, 16, 61, 1]
     tables = torch.zeros(B, 4, dtype=torch.int32)
     page = 0
@@ -176,9 +176,11 @@ def test_kv_append_and_paged_decode():
     v = torch.cat(all_v)
     slots = torch.tensor(all_slots, dtype=torch.long, device=DEV)
     ops.kv_append(k, v, slots, kc, vc)
-    # verify append round-trip
+    # verify append round-trip (K token-major, V d-major)
     kc_flat = kc.view(-1, Hkv, D)
     assert torch.equal(kc_flat[slots].cpu(), k.cpu())
+    vc_flat = vc.permute(0, 3, 1, 2).reshape(-1, Hkv, D)
+    assert torch.equal(vc_flat[slots].cpu(), v.cpu())
     q = _mt(B, Hq, D, seed=7)
     sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
     out = ops.paged_attn_decode(q, kc, vc, tables.to(DEV), sl, scale=D ** -0.5)
