@@ -88,8 +88,10 @@ def main():
     dev = tr.device
     def one_update(u):
         ro, gs = tr._rollout(u)
-        td = tr.algo.make_train_data(tr, ro, gs)
-        tr._update(td)
+        with tr.timers.phase("score"):
+            td = tr.algo.make_train_data(tr, ro, gs)
+        with tr.timers.phase("update"):
+            tr._update(td)
 
     # quiet the per-update metric machinery during timing
     for w in range(args.warmup):

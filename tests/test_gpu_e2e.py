@@ -24,26 +24,29 @@ def test_sampler_gpu_greedy_matches_oracle():
     params = SamplingParams(n=1, temperature=0.0, top_p=1.0, max_tokens=12, seed=1)
     out = eng.generate(prompts, params)
 
-    def oracle(prompt, steps):
+    def oracle_follow(prompt, got, steps):
+        """Replay the ENGINE's tokens through full recompute; at each step
+        report the oracle argmax and the margin between the engine's token
+        and the best token.  A bookkeeping bug shows up as disagreement at a
+        CONFIDENT step; bf16 tie-flips only at near-zero margins."""
         toks = list(prompt)
-        for _ in range(steps):
+        report = []
+        for j in range(steps):
             ids, cu, mx, pos = pack_sequences([torch.tensor(toks)], device=DEV)
             h = m(ids, CausalLM.train_ctx(cu, mx, pos))
-            toks.append(int(m.logits(h[-1:]).argmax()))
-        return toks[len(prompt):]
+            logits = m.logits(h[-1:]).float()[0]
+            best = int(logits.argmax())
+            margin = float(logits[best] - logits[got[j]])
+            report.append((got[j], best, margin))
+            toks.append(got[j])  # follow the engine's trajectory
+        return report
 
     for i, p in enumerate(prompts):
-        want = oracle(p, 12)
         got = out[i].tolist()
-        # bf16 paged decode vs full recompute legitimately differ in rounding;
-        # with random-init (near-flat) logits a near-tie can flip a token and
-        # derail the continuation.  Require agreement on the first tokens.
-        agree = 0
-        for a, b in zip(got, want):
-            if a != b:
-                break
-            agree += 1
-        assert agree >= 4, (i, got, want)
+        rep = oracle_follow(p, got, 12)
+        for j, (g, best, margin) in enumerate(rep):
+            # engine's pick must be the oracle's argmax, or within bf16 noise
+            assert g == best or margin < 0.08, (i, j, rep)
 
 
 def test_grpo_gpu_short_training():
