@@ -66,6 +66,7 @@ def main():
         kl_coef=0.05, learning_rate=3e-6,
         gradient_checkpointing=True,
         score_token_budget=65536,
+        train_token_budget=49152,
         output_dir=os.environ.get("BENCH_OUT", "/tmp/nanorlhf_bench"),
         save_steps=0, log_samples=0, report_to="none",
         missing_eos_penalty=1.0,

@@ -223,7 +223,7 @@ __global__ void fa_bwd_preprocess_kernel(const short* __restrict__ dout,
 
 // =========================================================== BACKWARD ====
 template <int D, bool CAUSAL>
-__global__ void fa_bwd_kernel(const short* __restrict__ dout,
+__global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict__ dout,
                               const short* __restrict__ q,
                               const short* __restrict__ k,
                               const short* __restrict__ v,

@@ -46,14 +46,17 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
 
 // ------------------------------------------------------------- RMSNorm bwd
 // dx = inv * (g - x_hat * mean(g * x_hat)),  g = dy*w,  x_hat = x*inv
-// dw accumulated fp32 via atomics (rows >> H).
+// dw accumulated fp32 via atomics into NSLAB row-spread slabs (a single
+// [H] target serializes: profiled 145 us/call from atomic contention);
+// the host sums the slabs.
+#define RMS_NSLAB 64
 template <int BLOCK>
 __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
                                    const short* __restrict__ x,
                                    const short* __restrict__ w,
                                    const float* __restrict__ invrms,
                                    short* __restrict__ dx,
-                                   float* __restrict__ dw,
+                                   float* __restrict__ dw,  // [RMS_NSLAB, H]
                                    int H) {
   __shared__ float scratch[BLOCK / 64];
   const long row = blockIdx.x;
@@ -83,8 +86,9 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
     for (int j = 0; j < 8; j++) {
       float xh = bf2f(xv[j]) * inv;
       float g = bf2f(dv[j]) * bf2f(wv[j]);
-      o[j] = f2bf(inv == 0.f ? 0.f : (g - xh * dot) * inv);
-      atomicAdd(&dw[i * 8 + j], bf2f(dv[j]) * bf2f(f2bf(xh)));
+      o[j] = f2bf((g - xh * dot) * inv);
+      atomicAdd(&dw[(row % RMS_NSLAB) * (long)H + i * 8 + j],
+                bf2f(dv[j]) * bf2f(f2bf(xh)));
     }
     *reinterpret_cast<s16x8*>(dxr + i * 8) = o;
   }
@@ -203,14 +207,14 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int H = x.size(-1);
   const long N = x.numel() / H;
   auto dx = torch::empty_like(x);
-  auto dwf = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto dwf = torch::zeros({64, H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL((rmsnorm_bwd_kernel<256>), dim3(N), dim3(256), 0, stream,
                      (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
                      (const short*)w.data_ptr(), invrms.data_ptr<float>(),
                      (short*)dx.data_ptr(), dwf.data_ptr<float>(), H);
   HIP_CHECK_LAST();
-  return {dx, dwf.to(torch::kBFloat16)};
+  return {dx, dwf.sum(0).to(torch::kBFloat16)};
 }
 
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions,
