@@ -19,6 +19,8 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor
 // sampling.hip
 torch::Tensor sample_topp(torch::Tensor logits, double temperature, double top_p,
                           long seed, long step);
+torch::Tensor sample_topp_dev(torch::Tensor logits, double temperature, double top_p,
+                              long seed, torch::Tensor step);
 // kvcache.hip
 void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor slots,
                torch::Tensor k_cache, torch::Tensor v_cache);
@@ -45,6 +47,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_backward_dlogits", &ce_backward_dlogits);
   m.def("adamw_step", &adamw_step);
   m.def("sample_topp", &sample_topp);
+  m.def("sample_topp_dev", &sample_topp_dev);
   m.def("kv_append", &kv_append);
   m.def("paged_attn_decode", &paged_attn_decode);
   m.def("fa_fwd_varlen", &fa_fwd_varlen);

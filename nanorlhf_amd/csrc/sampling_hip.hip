@@ -18,7 +18,10 @@
 __global__ void sample_topp_kernel(const short* __restrict__ logits,
                                    long* __restrict__ out,
                                    int V, float temperature, float top_p,
-                                   unsigned long long seed, unsigned long long step) {
+                                   unsigned long long seed,
+                                   unsigned long long step_imm,
+                                   const long* __restrict__ step_ptr) {
+  const unsigned long long step = step_ptr ? (unsigned long long)*step_ptr : step_imm;
   __shared__ float red[SBLOCK / 64];
   __shared__ float hist[NBINS];
   __shared__ int argmax_sh;
@@ -207,7 +210,27 @@ torch::Tensor sample_topp(torch::Tensor logits, double temperature, double top_p
   hipLaunchKernelGGL(sample_topp_kernel, dim3(B), dim3(SBLOCK), 0, stream,
                      (const short*)logits.data_ptr(), out.data_ptr<long>(), V,
                      (float)temperature, (float)top_p,
-                     (unsigned long long)seed, (unsigned long long)step);
+                     (unsigned long long)seed, (unsigned long long)step, nullptr);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+torch::Tensor sample_topp_dev(torch::Tensor logits, double temperature, double top_p,
+                              long seed, torch::Tensor step) {
+  // step: int64 [1] device tensor, read inside the kernel — safe under
+  // hipGraph capture/replay (the immediate-arg form would freeze the step)
+  TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(step.scalar_type() == torch::kLong && step.is_cuda());
+  const long B = logits.size(0);
+  const int V = logits.size(1);
+  auto out = torch::empty({B}, logits.options().dtype(torch::kLong));
+  if (B == 0) return out;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(sample_topp_kernel, dim3(B), dim3(SBLOCK), 0, stream,
+                     (const short*)logits.data_ptr(), out.data_ptr<long>(), V,
+                     (float)temperature, (float)top_p,
+                     (unsigned long long)seed, 0ull, step.data_ptr<long>());
   HIP_CHECK_LAST();
   return out;
 }

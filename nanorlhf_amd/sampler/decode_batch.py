@@ -29,6 +29,7 @@ class DecodeBatch:
         self.last_tok = torch.zeros(0, dtype=torch.long, device=device)
         self.finished = torch.zeros(0, dtype=torch.bool, device=device)
         self.out = torch.zeros(0, max_tokens, dtype=torch.long, device=device)
+        self.version = 0  # bumped whenever state tensors are rebuilt
         if seqs:
             self.extend(seqs)
 
@@ -70,6 +71,7 @@ class DecodeBatch:
         self.out = torch.cat([self.out, out])
         self.gen_count = torch.cat([self.gen_count, gen])
         self.seqs.extend(seqs)
+        self.version += 1
 
     # ------------------------------------------------------------------ step
     def step_inputs(self):
@@ -82,18 +84,21 @@ class DecodeBatch:
         return self.last_tok, pos, slots, self.seq_lens, self.bt
 
     def commit(self, tokens: torch.Tensor, pad_token_id: int):
-        """Record sampled tokens; advance lengths; no host sync."""
+        """Record sampled tokens; advance lengths.  Fully IN-PLACE on the
+        batch's state tensors so the whole step is hipGraph-capturable (no
+        host sync, no tensor reassignment)."""
         write = (~self.finished) & (self.gen_count < self.max_tokens)
         idx = self.gen_count.clamp(max=self.max_tokens - 1).long().unsqueeze(1)
         cur = self.out.gather(1, idx).squeeze(1)
         val = torch.where(write, tokens, cur)
         self.out.scatter_(1, idx, val.unsqueeze(1))
-        self.gen_count = self.gen_count + write.int()
-        self.seq_lens = self.seq_lens + write.int()
-        self.last_tok = torch.where(write, tokens, self.last_tok)
+        wi = write.int()
+        self.gen_count.add_(wi)
+        self.seq_lens.add_(wi)
+        torch.where(write, tokens, self.last_tok, out=self.last_tok)
         if self.stop is not None:
-            self.finished = self.finished | (write & (tokens == self.stop))
-        self.finished = self.finished | (self.gen_count >= self.max_tokens)
+            self.finished.logical_or_(write & (tokens == self.stop))
+        self.finished.logical_or_(self.gen_count >= self.max_tokens)
 
     # ------------------------------------------------------------------ sync
     def compact(self):
@@ -126,6 +131,7 @@ class DecodeBatch:
         self.last_tok = self.last_tok[km]
         self.finished = self.finished[km]
         self.out = self.out[km]
+        self.version += 1
         return retired
 
     def all_finished(self) -> bool:

@@ -38,7 +38,7 @@ class SamplingParams:
 class SamplerEngine:
     def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
                  max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072,
-                 compact_interval: int = 16):
+                 compact_interval: int = 16, use_graphs: bool = True):
         self.model = model
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
@@ -47,6 +47,11 @@ class SamplerEngine:
         self.max_num_seqs = max_num_seqs
         self.prefill_chunk_tokens = prefill_chunk_tokens
         self.compact_interval = compact_interval
+        self.use_graphs = use_graphs and self.device.type == "cuda"
+        self._graph = None
+        self._graph_version = None
+        self._step_dev = (torch.zeros(1, dtype=torch.long, device=self.device)
+                          if self.device.type == "cuda" else None)
 
     # ----------------------------------------------------------------- utils
     def _slots_for_range(self, seq: SeqState, start: int, end: int) -> list[int]:
@@ -56,6 +61,13 @@ class SamplerEngine:
     def _sample_from_hidden(self, hidden_last: torch.Tensor,
                             params: SamplingParams) -> torch.Tensor:
         logits = self.model.logits(hidden_last)
+        if self._step_dev is not None:
+            # device-side step counter: correct under hipGraph replay
+            self._step_dev.add_(1)
+            return ops.ext().sample_topp_dev(logits.contiguous(),
+                                             float(params.temperature),
+                                             float(params.top_p),
+                                             int(params.seed), self._step_dev)
         self._sample_step += 1
         return ops.sample_tokens(logits, params.temperature, params.top_p,
                                  params.seed, self._sample_step)
@@ -93,6 +105,43 @@ class SamplerEngine:
                 if params.stop_token_id is not None and int(t) == params.stop_token_id:
                     s.finished = True
 
+    # --------------------------------------------------------------- decode
+    def _decode_step_eager(self, db: DecodeBatch, params: SamplingParams,
+                           pad_token_id: int):
+        ids, pos, slots, seq_lens, bt = db.step_inputs()
+        ctx = AttnContext(mode="decode", positions=pos,
+                          kv_caches=self.pool.layers, slots=slots,
+                          block_tables=bt, seq_lens=seq_lens)
+        hidden = self.model(ids, ctx)
+        tokens = self._sample_from_hidden(hidden, params)
+        db.commit(tokens, pad_token_id)
+
+    def _run_decode_step(self, db: DecodeBatch, params: SamplingParams,
+                         pad_token_id: int) -> int:
+        """One decode step; hipGraph-captured and replayed when the batch
+        state is stable.  Returns the number of steps actually taken (capture
+        does 2 warmup steps)."""
+        if not self.use_graphs:
+            self._decode_step_eager(db, params, pad_token_id)
+            return 1
+        if self._graph is None or self._graph_version != db.version:
+            torch.cuda.synchronize()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    self._decode_step_eager(db, params, pad_token_id)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._decode_step_eager(db, params, pad_token_id)
+            self._graph = g
+            self._graph_version = db.version
+            return 2  # the warmup steps ran; captured step executes on replay
+        self._graph.replay()
+        return 1
+
     # -------------------------------------------------------------- generate
     @torch.no_grad()
     def generate(self, prompts: list[list[int]], params: SamplingParams,
@@ -117,6 +166,9 @@ class SamplerEngine:
                     uid += 1
             done: list[SeqState] = []
             self._sample_step = 0
+            if self._step_dev is not None:
+                self._step_dev.zero_()
+            self._graph = None  # new rollout -> new capture
             db = DecodeBatch(self.pool, [], params.max_tokens,
                              params.stop_token_id, self.device)
             outer_guard = 0
@@ -145,14 +197,7 @@ class SamplerEngine:
                 # device-resident decode: no host sync inside the chunk
                 inner = 0
                 while len(db) and inner < self.compact_interval:
-                    ids, pos, slots, seq_lens, bt = db.step_inputs()
-                    ctx = AttnContext(mode="decode", positions=pos,
-                                      kv_caches=self.pool.layers, slots=slots,
-                                      block_tables=bt, seq_lens=seq_lens)
-                    hidden = self.model(ids, ctx)
-                    tokens = self._sample_from_hidden(hidden, params)
-                    db.commit(tokens, pad_token_id)
-                    inner += 1
+                    inner += self._run_decode_step(db, params, pad_token_id)
                 done.extend(db.compact())
                 outer_guard += 1
                 if outer_guard > 8 * (params.max_tokens // self.compact_interval + 2) \
