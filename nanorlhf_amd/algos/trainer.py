@@ -172,6 +172,9 @@ class RLHFTrainer:
         self.timers = PhaseTimers(sync_cuda=device.type == "cuda")
         self.global_step = 0
         self.episode = 0
+        # optional evaluation hook: eval_fn(trainer) -> dict of metrics
+        # (r1 mode's greedy MATH accuracy pass, grpo_r1_trainer.py:471-473,824-825)
+        self.eval_fn = None
 
     # ------------------------------------------------------------------ utils
     def _auto_pool_tokens(self, prompts) -> int:
@@ -428,6 +431,9 @@ class RLHFTrainer:
         n_updates = num_updates or self.sizes["num_updates"]
         # sync a run timestamp across ranks (reference broadcast, :241-242)
         run_tag = pdist.broadcast_scalar(time.time(), device=pdist.metric_device())
+        if self.eval_fn is not None and cfg.eval_at_start:
+            ev = {f"initial_{k}": v for k, v in self.eval_fn(self).items()}
+            self.logger.log(ev, 0)
         for update in range(1, n_updates + 1):
             t0 = time.time()
             ro, greedy_scores = self._rollout(update)
@@ -439,6 +445,11 @@ class RLHFTrainer:
             self.episode += self.sizes["batch_size"]
             dt = time.time() - t0
             self._log_update(ro, td, upd_stats, dt)
+            if (self.eval_fn is not None and cfg.eval_steps
+                    and self.global_step % cfg.eval_steps == 0):
+                ev = {f"eval_{k}_new": v for k, v in self.eval_fn(self).items()}
+                self.logger.log(ev, self.global_step)
+                self._last_metrics.update(ev)
             if cfg.save_steps and (self.global_step % cfg.save_steps == 0):
                 self.save()
         return self
@@ -481,6 +492,31 @@ class RLHFTrainer:
         self._last_metrics = m
 
     # ------------------------------------------------------------- checkpoint
+    def load_checkpoint(self, ckpt_dir: str):
+        """Resume policy adapter + optimizer + RNG from a checkpoint dir
+        (an improvement over the reference, whose custom train() never wires
+        resume_from_checkpoint — SURVEY.md §5 Checkpoint/resume)."""
+        import json
+        import os
+        from ..utils.checkpoint import CheckpointManager, load_rng_state
+        state = CheckpointManager.load_policy_state(ckpt_dir)
+        missing, unexpected = self.policy.load_state_dict(state, strict=False)
+        if unexpected:
+            raise RuntimeError(f"unexpected keys in checkpoint: {unexpected[:5]}")
+        opt_path = os.path.join(ckpt_dir, "optimizer.pt")
+        if os.path.exists(opt_path):
+            self.optimizer.load_state_dict(torch.load(opt_path, weights_only=False))
+        if os.path.exists(os.path.join(ckpt_dir, "rng_state.pth")):
+            load_rng_state(ckpt_dir)
+        with open(os.path.join(ckpt_dir, "trainer_state.json")) as f:
+            ts = json.load(f)
+        self.global_step = ts["global_step"]
+        self.episode = ts["episode"]
+        vm_path = os.path.join(ckpt_dir, "value_model", "pytorch_model.bin")
+        if self.value_model is not None and os.path.exists(vm_path):
+            self.value_model.load_state_dict(torch.load(vm_path, weights_only=True))
+        return self
+
     def save(self) -> str | None:
         if self.rank != 0:
             return None

@@ -82,6 +82,10 @@ class RLHFConfig:
     # ---- sparse-GRPO / r1 mode (grpo_r1_trainer.py) --------------------------
     sparse_filter: bool = False            # drop score==0 samples (:565-568)
 
+    # ---- eval (r1 mode: greedy accuracy pass, grpo_r1_trainer.py:824-825) ----
+    eval_steps: int = 0                    # 0 → no periodic eval
+    eval_at_start: bool = True             # initial accuracy (:471-473)
+
     # ---- checkpoint / logging ------------------------------------------------
     save_steps: int = 0                    # 0 → no periodic save (ref default 1)
     save_total_limit: int = 8

@@ -19,7 +19,7 @@ def constant_reward(texts_or_ids, value: float = 1.0) -> torch.Tensor:
 class MathRuleReward:
     def __init__(self, gold_answers: dict[str, str] | list[str],
                  correct: float = 1.0, incorrect: float = 0.0,
-                 timeout_s: float = 0.5):
+                 timeout_s: float = 0.5, require_boxed: bool = True):
         """gold_answers: mapping prompt-key → gold answer string, or a list
         aligned with the rollout order (the reference hashes the train set's
         questions → answers, grpo_r1.py:237-240)."""
@@ -27,6 +27,9 @@ class MathRuleReward:
         self.correct = correct
         self.incorrect = incorrect
         self.timeout_s = timeout_s
+        # require_boxed=False falls back to the LAST number in the text
+        # (the reference's answer_extraction.py also has non-boxed extractors)
+        self.require_boxed = require_boxed
 
     def _gold_for(self, i: int, key: str | None) -> str | None:
         if isinstance(self.gold, dict):
@@ -42,6 +45,10 @@ class MathRuleReward:
             if gold is None:
                 continue
             pred = extract_boxed(text)
+            if pred is None and not self.require_boxed:
+                import re
+                nums = re.findall(r"-?\d+(?:\.\d+)?", text)
+                pred = nums[-1] if nums else None
             if pred is None:
                 continue
             if answers_equal(pred, gold, self.timeout_s):
