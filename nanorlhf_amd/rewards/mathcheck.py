@@ -129,7 +129,8 @@ def answers_equal(pred: str, gold: str, sympy_timeout_s: float = 0.5) -> bool:
     na, nb = _num(a), _num(b)
     if na is not None and nb is not None:
         return abs(na - nb) <= 1e-6 * max(1.0, abs(nb))
-    if na is not None or nb is not None:
-        # one numeric, one symbolic → let sympy decide
-        pass
+    # warm sympy in the PARENT: the timeout-guarded child is a fork and must
+    # not pay the multi-second sympy import inside its budget
+    import sympy  # noqa: F401
+    import sympy.parsing.sympy_parser  # noqa: F401
     return call_with_timeout(_sympy_worker, (a, b), sympy_timeout_s)
