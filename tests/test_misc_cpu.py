@@ -80,3 +80,39 @@ def test_hh_shaped_prompts_reproducible():
     b = hh_shaped_prompts(5, 1000, seed=3)
     assert a == b
     assert all(2 <= t < 1000 for p in a for t in p)
+
+
+def test_checkpoint_resume_roundtrip(tmp_path):
+    """trainer.load_checkpoint restores policy + optimizer + step counters."""
+    import torch
+    from nanorlhf_amd.algos import reinforce
+    from nanorlhf_amd.algos.reinforce import ReinforceConfig
+    from nanorlhf_amd.data import hh_shaped_prompts
+    from nanorlhf_amd.models import CausalLM
+    from nanorlhf_amd.rewards import constant_reward
+
+    def mk(out):
+        cfg = ReinforceConfig(model_preset="tiny", dtype="float32", use_lora=True,
+                              lora_r=4, lora_alpha=8, per_device_train_batch_size=2,
+                              gradient_accumulation_steps=1, num_mini_batches=2,
+                              total_episodes=8, response_length=4, temperature=1.0,
+                              stop_token_id=1, output_dir=out, save_steps=1,
+                              gradient_checkpointing=False, score_token_budget=256)
+        torch.manual_seed(0)
+        policy = CausalLM.from_preset("tiny")
+        ref = CausalLM.from_preset("tiny")
+        ref.load_state_dict(policy.state_dict())
+        prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8)
+        return reinforce.make_trainer(cfg, policy, ref,
+                                      lambda s: constant_reward(s), prompts)
+
+    t1 = mk(str(tmp_path / "a"))
+    t1.train(num_updates=1)
+    ck = str(tmp_path / "a" / "checkpoint-1")
+    t2 = mk(str(tmp_path / "b"))
+    t2.load_checkpoint(ck)
+    assert t2.global_step == t1.global_step and t2.episode == t1.episode
+    for (n1, p1), (n2, p2) in zip(
+            ((n, p) for n, p in t1.policy.named_parameters() if p.requires_grad),
+            ((n, p) for n, p in t2.policy.named_parameters() if p.requires_grad)):
+        assert n1 == n2 and torch.allclose(p1, p2, atol=1e-7), n1
