@@ -37,6 +37,16 @@ DEVINL f32x16_t mfma32x32x16(bf16x8 a, bf16x8 b, f32x16_t c) {
 
 DEVINL void lds_fence_wave() { lds_fence_wave_kv(); }
 
+// XOR-swizzled LDS element index for a [row][bytes-per-row] bf16 tile:
+// spreads each 16-B slot across bank columns (row-major D=128 tiles read
+// with ds_read_b128 are otherwise up to 16-way bank conflicts —
+// cdna_hip_programming.md §6 G4).  ROWBYTES must be a power of two >= 32.
+template <int ROWBYTES>
+DEVINL int swz_idx(int row, int byte_in_row) {
+  constexpr int MASK = (ROWBYTES / 16) - 1;
+  return row * (ROWBYTES / 2) + ((byte_in_row ^ ((row & MASK) << 4)) >> 1);
+}
+
 DEVINL __bf16 f2bf16t(float f) {
   union { short s; __bf16 b; } u;
   u.s = f2bf(f);
@@ -107,9 +117,10 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
         kk = *reinterpret_cast<const s16x8*>(k + b);
         vv = *reinterpret_cast<const s16x8*>(v + b);
       }
-      *reinterpret_cast<s16x8*>(&Kt[tok * D + d0]) = kk;
+      *reinterpret_cast<s16x8*>(&Kt[swz_idx<2 * D>(tok, d0 * 2)]) = kk;
 #pragma unroll
-      for (int j = 0; j < 8; j++) Vt[(d0 + j) * KTILE + tok] = vv[j];
+      for (int j = 0; j < 8; j++)
+        Vt[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = vv[j];
     }
     __syncthreads();
     if (!wave_active) continue;
@@ -121,7 +132,8 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
       sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        s16x8 raw = *reinterpret_cast<const s16x8*>(&Kt[(16 * n + lo) * D + 32 * c + 8 * hi]);
+        s16x8 raw = *reinterpret_cast<const s16x8*>(
+            &Kt[swz_idx<2 * D>(16 * n + lo, 64 * c + 16 * hi)]);
         sc[n] = mfma16x16x32(qf[c], *reinterpret_cast<bf16x8*>(&raw), sc[n]);
       }
     }
@@ -186,7 +198,8 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
     }
 #pragma unroll
     for (int t = 0; t < D / 16; t++) {
-      s16x8 raw = *reinterpret_cast<const s16x8*>(&Vt[(t * 16 + lo) * KTILE + 8 * hi]);
+      s16x8 raw = *reinterpret_cast<const s16x8*>(
+          &Vt[swz_idx<2 * KTILE>(t * 16 + lo, 16 * hi)]);
       acc_o[t] = mfma16x16x32(pf, *reinterpret_cast<bf16x8*>(&raw), acc_o[t]);
     }
   }
@@ -269,10 +282,11 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
       kk = *reinterpret_cast<const s16x8*>(k + b);
       vv = *reinterpret_cast<const s16x8*>(v + b);
     }
-    *reinterpret_cast<s16x8*>(&Kt[tok * D + d0]) = kk;
-    *reinterpret_cast<s16x8*>(&Vt[tok * D + d0]) = vv;
+    *reinterpret_cast<s16x8*>(&Kt[swz_idx<2 * D>(tok, d0 * 2)]) = kk;
+    *reinterpret_cast<s16x8*>(&Vt[swz_idx<2 * D>(tok, d0 * 2)]) = vv;
 #pragma unroll
-    for (int j = 0; j < 8; j++) KT[(d0 + j) * KTILE + tok] = kk[j];
+    for (int j = 0; j < 8; j++)
+      KT[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = kk[j];
   }
   __syncthreads();
 
@@ -311,8 +325,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
         dofr[c] = *reinterpret_cast<bf16x8*>(&rd);
 #pragma unroll
         for (int j = 0; j < 8; j++) {
-          QT[wid][(32 * c + 8 * hi + j) * 16 + lo] = rq[j];
-          dOT[wid][(32 * c + 8 * hi + j) * 16 + lo] = rd[j];
+          QT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rq[j];
+          dOT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rd[j];
         }
       }
     }
@@ -325,8 +339,10 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
       dp[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        s16x8 rk = *reinterpret_cast<const s16x8*>(&Kt[(16 * n + lo) * D + 32 * c + 8 * hi]);
-        s16x8 rv = *reinterpret_cast<const s16x8*>(&Vt[(16 * n + lo) * D + 32 * c + 8 * hi]);
+        s16x8 rk = *reinterpret_cast<const s16x8*>(
+            &Kt[swz_idx<2 * D>(16 * n + lo, 64 * c + 16 * hi)]);
+        s16x8 rv = *reinterpret_cast<const s16x8*>(
+            &Vt[swz_idx<2 * D>(16 * n + lo, 64 * c + 16 * hi)]);
         sc[n] = mfma16x16x32(qfr[c], *reinterpret_cast<bf16x8*>(&rk), sc[n]);
         dp[n] = mfma16x16x32(dofr[c], *reinterpret_cast<bf16x8*>(&rv), dp[n]);
       }
@@ -361,8 +377,10 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
       bf16x8 sA = *reinterpret_cast<bf16x8*>(&rs);
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        s16x8 rdo = *reinterpret_cast<const s16x8*>(&dOT[wid][(32 * c + lo5) * 16 + 8 * hi5]);
-        s16x8 rqt = *reinterpret_cast<const s16x8*>(&QT[wid][(32 * c + lo5) * 16 + 8 * hi5]);
+        s16x8 rdo = *reinterpret_cast<const s16x8*>(
+            &dOT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
+        s16x8 rqt = *reinterpret_cast<const s16x8*>(
+            &QT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
         acc_dv[c] = mfma32x32x16(pA, *reinterpret_cast<bf16x8*>(&rdo), acc_dv[c]);
         acc_dk[c] = mfma32x32x16(sA, *reinterpret_cast<bf16x8*>(&rqt), acc_dk[c]);
       }
@@ -374,7 +392,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
       bf16x8 dsA = *reinterpret_cast<bf16x8*>(&rds);
 #pragma unroll
       for (int t = 0; t < D / 16; t++) {
-        s16x8 rkt = *reinterpret_cast<const s16x8*>(&KT[(t * 16 + lo) * KTILE + 8 * hi]);
+        s16x8 rkt = *reinterpret_cast<const s16x8*>(
+            &KT[swz_idx<2 * KTILE>(t * 16 + lo, 16 * hi)]);
         f32x4 dq = mfma16x16x32(dsA, *reinterpret_cast<bf16x8*>(&rkt), f32x4{0.f, 0.f, 0.f, 0.f});
 #pragma unroll
         for (int r = 0; r < 4; r++) {

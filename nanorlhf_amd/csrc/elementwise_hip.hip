@@ -46,51 +46,65 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
 
 // ------------------------------------------------------------- RMSNorm bwd
 // dx = inv * (g - x_hat * mean(g * x_hat)),  g = dy*w,  x_hat = x*inv
-// dw accumulated fp32 via atomics into NSLAB row-spread slabs (a single
-// [H] target serializes: profiled 145 us/call from atomic contention);
-// the host sums the slabs.
-#define RMS_NSLAB 64
+// Each block processes ROWS_PER_BLOCK rows; every thread owns fixed column
+// slices, accumulating its dw partial in REGISTERS across the block's rows
+// (no per-element atomics — the atomic version measured 1.0 ms/call from
+// contention), then one global atomicAdd per element per block.
 template <int BLOCK>
 __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
                                    const short* __restrict__ x,
                                    const short* __restrict__ w,
                                    const float* __restrict__ invrms,
                                    short* __restrict__ dx,
-                                   float* __restrict__ dw,  // [RMS_NSLAB, H]
-                                   int H) {
+                                   float* __restrict__ dw,  // [H]
+                                   long N, int H, int rows_per_block) {
   __shared__ float scratch[BLOCK / 64];
-  const long row = blockIdx.x;
-  const short* dyr = dy + row * (long)H;
-  const short* xr = x + row * (long)H;
-  short* dxr = dx + row * (long)H;
-  const float inv = invrms[row];
+  constexpr int MAX_SLICES = 4;  // supports H up to BLOCK*8*4 = 8192
   const int nvec = H / 8;
-  float dot = 0.f;
-  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
-    s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i * 8);
-    s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
-    s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
+  float dwacc[MAX_SLICES][8];
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      float g = bf2f(dv[j]) * bf2f(wv[j]);
-      dot += g * bf2f(xv[j]) * inv;
+  for (int sl = 0; sl < MAX_SLICES; sl++)
+#pragma unroll
+    for (int j = 0; j < 8; j++) dwacc[sl][j] = 0.f;
+
+  const long row0 = (long)blockIdx.x * rows_per_block;
+  const long row1 = min(row0 + rows_per_block, N);
+  for (long row = row0; row < row1; row++) {
+    const short* dyr = dy + row * (long)H;
+    const short* xr = x + row * (long)H;
+    short* dxr = dx + row * (long)H;
+    const float inv = invrms[row];
+    float dot = 0.f;
+    for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i * 8);
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        dot += bf2f(dv[j]) * bf2f(wv[j]) * bf2f(xv[j]) * inv;
+    }
+    dot = block_sum<BLOCK>(dot, scratch) / (float)H;
+    int sl = 0;
+    for (int i = threadIdx.x; i < nvec; i += BLOCK, sl++) {
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i * 8);
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float xh = bf2f(xv[j]) * inv;
+        const float g = bf2f(dv[j]) * bf2f(wv[j]);
+        o[j] = f2bf((g - xh * dot) * inv);
+        if (sl < MAX_SLICES) dwacc[sl][j] += bf2f(dv[j]) * bf2f(f2bf(xh));
+      }
+      *reinterpret_cast<s16x8*>(dxr + i * 8) = o;
     }
   }
-  dot = block_sum<BLOCK>(dot, scratch) / (float)H;
-  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
-    s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i * 8);
-    s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
-    s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
-    s16x8 o;
+  // drain register partials: one atomic per element per block
+  int sl = 0;
+  for (int i = threadIdx.x; i < nvec && sl < MAX_SLICES; i += BLOCK, sl++) {
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      float xh = bf2f(xv[j]) * inv;
-      float g = bf2f(dv[j]) * bf2f(wv[j]);
-      o[j] = f2bf((g - xh * dot) * inv);
-      atomicAdd(&dw[(row % RMS_NSLAB) * (long)H + i * 8 + j],
-                bf2f(dv[j]) * bf2f(f2bf(xh)));
-    }
-    *reinterpret_cast<s16x8*>(dxr + i * 8) = o;
+    for (int j = 0; j < 8; j++) atomicAdd(&dw[i * 8 + j], dwacc[sl][j]);
   }
 }
 
@@ -206,15 +220,20 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int H = x.size(-1);
   const long N = x.numel() / H;
+  TORCH_CHECK(H <= 256 * 8 * 4, "hidden too large for rmsnorm_bwd slices");
   auto dx = torch::empty_like(x);
-  auto dwf = torch::zeros({64, H}, x.options().dtype(torch::kFloat32));
+  auto dwf = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((rmsnorm_bwd_kernel<256>), dim3(N), dim3(256), 0, stream,
+  // ~4 blocks/CU worth of parallelism; more rows per block = fewer atomics
+  const int rows_per_block = (int)std::max<long>(1, (N + 1023) / 1024);
+  const long grid = (N + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL((rmsnorm_bwd_kernel<256>), dim3(grid), dim3(256), 0, stream,
                      (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
                      (const short*)w.data_ptr(), invrms.data_ptr<float>(),
-                     (short*)dx.data_ptr(), dwf.data_ptr<float>(), H);
+                     (short*)dx.data_ptr(), dwf.data_ptr<float>(), N, H,
+                     rows_per_block);
   HIP_CHECK_LAST();
-  return {dx, dwf.sum(0).to(torch::kBFloat16)};
+  return {dx, dwf.to(torch::kBFloat16)};
 }
 
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions,
