@@ -74,7 +74,11 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       float mx2 = __shfl_xor(lmax, off);
       int a2 = __shfl_xor(am, off);
       const float mn = fmaxf(m, m2);
-      s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+      // -inf guards: threads with no elements (V < 8*SBLOCK) hold m = -inf
+      // and exp(-inf - -inf) would poison the merge with NaN
+      const float c1 = (m == -INFINITY) ? 0.f : __expf(m - mn);
+      const float c2 = (m2 == -INFINITY) ? 0.f : __expf(m2 - mn);
+      s = s * c1 + s2 * c2;
       m = mn;
       if (mx2 > lmax || (mx2 == lmax && a2 < am)) { lmax = mx2; am = a2; }
     }
@@ -85,7 +89,9 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       int A = wa_[0];
       for (int w = 1; w < SBLOCK / 64; w++) {
         const float mn = fmaxf(M, sm[w]);
-        S = S * __expf(M - mn) + ss[w] * __expf(sm[w] - mn);
+        const float c1 = (M == -INFINITY) ? 0.f : __expf(M - mn);
+        const float c2 = (sm[w] == -INFINITY) ? 0.f : __expf(sm[w] - mn);
+        S = S * c1 + ss[w] * c2;
         M = mn;
         if (wm_[w] > MX || (wm_[w] == MX && wa_[w] < A)) { MX = wm_[w]; A = wa_[w]; }
       }
