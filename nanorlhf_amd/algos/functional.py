@@ -16,6 +16,15 @@ import torch
 INVALID_LOGPROB = 1.0
 
 
+def disable_dropout_in_model(model: torch.nn.Module) -> torch.nn.Module:
+    """Set every Dropout's p to 0 (trl helper imported at
+    grpo_trainer.py:58-71; our models only carry LoRA dropout)."""
+    for m in model.modules():
+        if isinstance(m, torch.nn.Dropout):
+            m.p = 0.0
+    return model
+
+
 def masked_mean(values: torch.Tensor, mask: torch.Tensor, axis=None) -> torch.Tensor:
     """Mean of `values` over positions where mask==1.
 

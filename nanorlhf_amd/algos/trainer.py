@@ -101,7 +101,8 @@ class RLHFTrainer:
                  ref_policy: Optional[CausalLM], reward_fn: Callable,
                  train_prompts: list[list[int]],
                  value_model: Optional[ScalarHeadModel] = None,
-                 device: Optional[torch.device] = None):
+                 device: Optional[torch.device] = None,
+                 callbacks: Optional[list] = None):
         self.cfg = cfg
         self.algo = algo
         self.rank, self.local_rank, self.world = pdist.init_distributed()
@@ -175,6 +176,10 @@ class RLHFTrainer:
         # optional evaluation hook: eval_fn(trainer) -> dict of metrics
         # (r1 mode's greedy MATH accuracy pass, grpo_r1_trainer.py:471-473,824-825)
         self.eval_fn = None
+        self.callbacks = list(callbacks or [])
+        self._plateau_best = None
+        self._plateau_bad = 0
+        self._lr_scale = 1.0
 
     # ------------------------------------------------------------------ utils
     def _auto_pool_tokens(self, prompts) -> int:
@@ -203,6 +208,9 @@ class RLHFTrainer:
             return cfg.learning_rate * (t + 1) / cfg.warmup_steps
         if cfg.lr_scheduler_type == "constant":
             return cfg.learning_rate
+        if cfg.lr_scheduler_type == "reduce_lr_on_plateau":
+            # reference PPO value path (ppo.py:97-98): halve on plateau
+            return cfg.learning_rate * self._lr_scale
         # cosine_with_min_lr (reference grpo.py:119-120)
         min_lr = cfg.learning_rate * cfg.min_lr_ratio
         prog = min(1.0, t / max(1, total))
@@ -434,24 +442,50 @@ class RLHFTrainer:
         if self.eval_fn is not None and cfg.eval_at_start:
             ev = {f"initial_{k}": v for k, v in self.eval_fn(self).items()}
             self.logger.log(ev, 0)
+        stop = False
         for update in range(1, n_updates + 1):
-            t0 = time.time()
-            ro, greedy_scores = self._rollout(update)
-            with self.timers.phase("score"):
-                td = self.algo.make_train_data(self, ro, greedy_scores)
-            with self.timers.phase("update"):
-                upd_stats = self._update(td)
+            try:
+                t0 = time.time()
+                ro, greedy_scores = self._rollout(update)
+                with self.timers.phase("score"):
+                    td = self.algo.make_train_data(self, ro, greedy_scores)
+                with self.timers.phase("update"):
+                    upd_stats = self._update(td)
+            except Exception:
+                # crash recovery: persist what we have before surfacing the
+                # error (the reference loses the run past the last save —
+                # SURVEY §5 "Failure detection: none")
+                if self.global_step > 0:
+                    self.save()
+                raise
             self.global_step += 1
             self.episode += self.sizes["batch_size"]
             dt = time.time() - t0
             self._log_update(ro, td, upd_stats, dt)
+            if self.cfg.lr_scheduler_type == "reduce_lr_on_plateau":
+                v = self._last_metrics.get("objective/rlhf_reward_old", 0.0)
+                if self._plateau_best is None or v > self._plateau_best:
+                    self._plateau_best = v
+                    self._plateau_bad = 0
+                else:
+                    self._plateau_bad += 1
+                    if self._plateau_bad >= self.cfg.plateau_patience:
+                        self._lr_scale *= self.cfg.plateau_factor
+                        self._plateau_bad = 0
+            for cb in self.callbacks:
+                if cb.on_update_end(self, self._last_metrics):
+                    stop = True
             if (self.eval_fn is not None and cfg.eval_steps
                     and self.global_step % cfg.eval_steps == 0):
                 ev = {f"eval_{k}_new": v for k, v in self.eval_fn(self).items()}
                 self.logger.log(ev, self.global_step)
                 self._last_metrics.update(ev)
             if cfg.save_steps and (self.global_step % cfg.save_steps == 0):
-                self.save()
+                ck = self.save()
+                for cb in self.callbacks:
+                    cb.on_save(self, ck)
+            if stop:
+                break
         return self
 
     # ---------------------------------------------------------------- logging

@@ -59,6 +59,8 @@ class RLHFConfig:
     learning_rate: float = 3e-6
     min_lr_ratio: float = 0.1          # cosine_with_min_lr (grpo.py:119-120)
     lr_scheduler_type: str = "cosine_with_min_lr"  # or "constant", "reduce_lr_on_plateau"
+    plateau_patience: int = 10       # reduce_lr_on_plateau (PPO/ppo.py:97-98)
+    plateau_factor: float = 0.5
     warmup_steps: int = 0
     weight_decay: float = 0.0
     adam_beta1: float = 0.9

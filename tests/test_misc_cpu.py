@@ -116,3 +116,30 @@ def test_checkpoint_resume_roundtrip(tmp_path):
             ((n, p) for n, p in t1.policy.named_parameters() if p.requires_grad),
             ((n, p) for n, p in t2.policy.named_parameters() if p.requires_grad)):
         assert n1 == n2 and torch.allclose(p1, p2, atol=1e-7), n1
+
+
+def test_python_executor_sandbox():
+    from nanorlhf_amd.rewards.python_exec import PythonExecutor
+    ex = PythonExecutor(timeout_s=3.0)
+    r = ex.run("answer = 6*7\nprint('hi')")
+    assert r["ok"] and r["answer"] == 42 and "hi" in r["stdout"]
+    r2 = ex.run("while True: pass")
+    assert not r2["ok"] and r2["error"] == "timeout"
+    r3 = ex.run("raise ValueError('x')")
+    assert not r3["ok"]
+
+
+def test_early_stopping_callback():
+    from nanorlhf_amd.utils.callbacks import EarlyStoppingCallback
+    cb = EarlyStoppingCallback(metric="m", patience=2, greater_is_better=True)
+    assert not cb.on_update_end(None, {"m": 1.0})
+    assert not cb.on_update_end(None, {"m": 0.9})
+    assert cb.on_update_end(None, {"m": 0.8})  # second bad step -> stop
+
+
+def test_disable_dropout():
+    import torch
+    from nanorlhf_amd.algos.functional import disable_dropout_in_model
+    m = torch.nn.Sequential(torch.nn.Linear(2, 2), torch.nn.Dropout(0.5))
+    disable_dropout_in_model(m)
+    assert m[1].p == 0.0
