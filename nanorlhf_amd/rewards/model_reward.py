@@ -15,11 +15,15 @@ from ..utils.offload import OffloadEngine
 
 class ModelReward:
     def __init__(self, model: ScalarHeadModel, device, token_budget: int = 65536,
-                 offload: OffloadEngine | None = None):
+                 offload: OffloadEngine | None = None, max_len: int = 512):
+        """max_len mirrors the reference RM's tokenizer truncation —
+        deberta-v3-large caps at 512 positions, so its rewards are computed
+        on the first 512 tokens of question+response (GRPO/grpo.py:180-192)."""
         self.model = model
         self.device = torch.device(device)
         self.token_budget = token_budget
         self.offload = offload
+        self.max_len = max_len
 
     @torch.no_grad()
     def __call__(self, sequences: list[list[int]]) -> torch.Tensor:
@@ -31,6 +35,8 @@ class ModelReward:
             self.offload.model_to_device(self.model)
             self.offload.synchronize()
         self.model.eval()
+        if self.max_len:
+            sequences = [s[: self.max_len] for s in sequences]
         scores = torch.empty(len(sequences), dtype=torch.float32)
         i = 0
         while i < len(sequences):
