@@ -64,7 +64,8 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
                               float* __restrict__ lse,
                               int Hq, int Hkv, float scale) {
   constexpr int QTILE = 64;   // per workgroup; 16 per wave
-  constexpr int KTILE = 32;
+  constexpr int KTILE = 64;   // KV tokens staged per iteration (4 subtiles)
+  constexpr int NSUB = KTILE / 16;
   constexpr int NC = D / 32;  // K-chunks per mfma row
   const int seq = blockIdx.y;
   const int h = blockIdx.z;
@@ -79,11 +80,11 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
   const int lo = lane & 15;
   const int qr0 = q0 + wid * 16;      // this wave's first q row (tile-local)
 
-  __shared__ short Kt[KTILE * D];       // row-major [tok][d]
-  __shared__ short Vt[D * KTILE];       // transposed [d][tok]
+  __shared__ short Kt[KTILE * D];       // row-major [tok][d], XOR-swizzled
+  __shared__ short Vt[D * KTILE];       // transposed [d][tok], XOR-swizzled
   __shared__ short Pb[4][16 * KTILE];   // per-wave P [q][kv]
 
-  // ---- Q fragments: lane holds Q[lo][32c + 8hi + j], scaled later -------
+  // ---- Q fragments: lane holds Q[lo][32c + 8hi + j] ---------------------
   bf16x8 qf[NC];
   const bool wave_active = qr0 < len;
   {
@@ -105,7 +106,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
 
   const int kv_end = CAUSAL ? min(len, q0 + QTILE) : len;
   for (int kv0 = 0; kv0 < kv_end; kv0 += KTILE) {
-    // ---- cooperative K/V stage ----------------------------------------
+    // ---- cooperative K/V stage (Kt row-major, Vt transposed) ----------
     __syncthreads();
     for (int idx = threadIdx.x; idx < KTILE * D / 8; idx += 256) {
       const int tok = idx / (D / 8);
@@ -125,10 +126,10 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
     __syncthreads();
     if (!wave_active) continue;
 
-    // ---- S = Q K^T : two 16-col subtiles -------------------------------
-    f32x4 sc[2];
+    // ---- S = Q K^T : NSUB 16-col subtiles ------------------------------
+    f32x4 sc[NSUB];
 #pragma unroll
-    for (int n = 0; n < 2; n++) {
+    for (int n = 0; n < NSUB; n++) {
       sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int c = 0; c < NC; c++) {
@@ -139,9 +140,9 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
     }
     // ---- mask + online softmax ----------------------------------------
     // C layout: row (q) = 4*hi + reg, col (kv) = lo + 16*n
-    float pm[2][4];
+    float pm[NSUB][4];
 #pragma unroll
-    for (int n = 0; n < 2; n++) {
+    for (int n = 0; n < NSUB; n++) {
 #pragma unroll
       for (int r = 0; r < 4; r++) {
         const int qi = qr0 + 4 * hi + r;
@@ -153,7 +154,9 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
     float rowmax[4];
 #pragma unroll
     for (int r = 0; r < 4; r++) {
-      float x = fmaxf(pm[0][r], pm[1][r]);
+      float x = pm[0][r];
+#pragma unroll
+      for (int n = 1; n < NSUB; n++) x = fmaxf(x, pm[n][r]);
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off));
       rowmax[r] = x;
@@ -168,7 +171,7 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
     // P = exp(score - m), row sums, stage P to per-wave LDS
     float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int n = 0; n < 2; n++) {
+    for (int n = 0; n < NSUB; n++) {
 #pragma unroll
       for (int r = 0; r < 4; r++) {
         const float p = (pm[n][r] == -INFINITY) ? 0.f : __expf(pm[n][r] - m[r]);
@@ -190,17 +193,22 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
       for (int r = 0; r < 4; r++) acc_o[t][r] *= fac[r];
     }
     lds_fence_wave();
-    // ---- PV: A = P (LDS), B = V^T slice (contiguous via Vt) ------------
-    bf16x8 pf;
+    // ---- PV: A = P (LDS), B = V^T slices; K=64 -> 2 mfma per d-tile ----
+    bf16x8 pf[2];
     {
-      s16x8 raw = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 8 * hi]);
-      pf = *reinterpret_cast<bf16x8*>(&raw);
+      s16x8 r0 = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 8 * hi]);
+      s16x8 r1 = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 32 + 8 * hi]);
+      pf[0] = *reinterpret_cast<bf16x8*>(&r0);
+      pf[1] = *reinterpret_cast<bf16x8*>(&r1);
     }
 #pragma unroll
     for (int t = 0; t < D / 16; t++) {
-      s16x8 raw = *reinterpret_cast<const s16x8*>(
+      s16x8 rv0 = *reinterpret_cast<const s16x8*>(
           &Vt[swz_idx<2 * KTILE>(t * 16 + lo, 16 * hi)]);
-      acc_o[t] = mfma16x16x32(pf, *reinterpret_cast<bf16x8*>(&raw), acc_o[t]);
+      acc_o[t] = mfma16x16x32(pf[0], *reinterpret_cast<bf16x8*>(&rv0), acc_o[t]);
+      s16x8 rv1 = *reinterpret_cast<const s16x8*>(
+          &Vt[swz_idx<2 * KTILE>(t * 16 + lo, 64 + 16 * hi)]);
+      acc_o[t] = mfma16x16x32(pf[1], *reinterpret_cast<bf16x8*>(&rv1), acc_o[t]);
     }
   }
 
