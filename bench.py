@@ -64,8 +64,9 @@ def main():
         sample_n=args.sample_n, response_length=args.response_length,
         temperature=0.7, top_p=0.95, stop_token_id=1, pad_token_id=0,
         kl_coef=0.05, learning_rate=3e-6,
-        gradient_checkpointing=True,
-        score_token_budget=65536,
+        # 288 GB HBM: no recompute needed at these batch shapes
+        gradient_checkpointing=False,
+        score_token_budget=98304,
         train_token_budget=49152,
         output_dir=os.environ.get("BENCH_OUT", "/tmp/nanorlhf_bench"),
         save_steps=0, log_samples=0, report_to="none",
