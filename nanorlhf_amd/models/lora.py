@@ -21,8 +21,7 @@ class LoraConfig:
     r: int = 64
     alpha: int = 16
     dropout: float = 0.0
-    target_modules: tuple = ("q_proj", "k_proj", "v_proj", "o_proj",
-                             "gate_up_proj", "down_proj")
+    target_modules: tuple = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj")
     # fully-trained modules (reference modules_to_save: embed_tokens, lm_head)
     modules_to_save: tuple = ("embed_tokens", "lm_head")
 

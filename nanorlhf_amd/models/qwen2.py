@@ -52,17 +52,22 @@ class Attention(nn.Module):
         self.num_kv_heads = cfg.num_kv_heads
         self.head_dim = cfg.head_dim
         H = cfg.hidden_size
-        self.q_proj = nn.Linear(H, cfg.num_heads * cfg.head_dim, bias=cfg.qkv_bias)
-        self.k_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=cfg.qkv_bias)
-        self.v_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=cfg.qkv_bias)
+        self.q_size = cfg.num_heads * cfg.head_dim
+        self.kv_size = cfg.num_kv_heads * cfg.head_dim
+        # fused qkv projection: one GEMM instead of three (MI355X-first —
+        # bigger N per launch; the reference's transformers Qwen2 keeps
+        # separate q/k/v linears)
+        self.qkv_proj = nn.Linear(H, self.q_size + 2 * self.kv_size, bias=cfg.qkv_bias)
         self.o_proj = nn.Linear(cfg.num_heads * cfg.head_dim, H, bias=False)
         self.scale = cfg.head_dim ** -0.5
 
     def forward(self, x: torch.Tensor, rope_table: torch.Tensor, ctx: AttnContext):
         T = x.shape[0]
-        q = self.q_proj(x).view(T, self.num_heads, self.head_dim)
-        k = self.k_proj(x).view(T, self.num_kv_heads, self.head_dim)
-        v = self.v_proj(x).view(T, self.num_kv_heads, self.head_dim)
+        qkv = self.qkv_proj(x)
+        q = qkv[:, : self.q_size].reshape(T, self.num_heads, self.head_dim)
+        k = qkv[:, self.q_size: self.q_size + self.kv_size].reshape(
+            T, self.num_kv_heads, self.head_dim)
+        v = qkv[:, self.q_size + self.kv_size:].reshape(T, self.num_kv_heads, self.head_dim)
         q = ops.rope_apply(q, rope_table, ctx.positions)
         k = ops.rope_apply(k, rope_table, ctx.positions)
 
