@@ -341,10 +341,29 @@ class RLHFTrainer:
                                  "vf_clipfrac", "entropy", "ratio", "ratio_var")}
         mb_rows_target = self.sizes["local_mini_batch_size"]
         lens = [len(p) + len(r) for p, r in zip(td.prompts, td.responses)]
+        # DP ranks MUST run the same number of minibatches: each minibatch
+        # boundary fires one bucketed all-reduce per parameter, so a rank
+        # with fewer rows (sparse-GRPO filtering is data-dependent,
+        # grpo_r1_trainer.py:565-568) would otherwise hang the others.
+        n_mb_local = max(1, (R + mb_rows_target - 1) // mb_rows_target)
+        n_mb = n_mb_local
+        if self.world > 1:
+            import torch.distributed as _dist
+            t = torch.tensor([n_mb_local], dtype=torch.long,
+                             device=pdist.metric_device())
+            _dist.all_reduce(t, op=_dist.ReduceOp.MAX)
+            n_mb = int(t.item())
         for epoch in range(cfg.num_ppo_epochs):
             perm = torch.randperm(R, generator=self._keep_gen).tolist()
-            for mb_start in range(0, R, mb_rows_target):
+            for mb_i in range(n_mb):
+                mb_start = mb_i * mb_rows_target
                 mb_rows = perm[mb_start: mb_start + mb_rows_target]
+                zero_weight = False
+                if not mb_rows:
+                    # surplus minibatch on this rank: run one row at zero
+                    # loss weight so gradient collectives stay matched
+                    mb_rows = [perm[0] if perm else 0]
+                    zero_weight = True
                 budget = cfg.train_token_budget
                 if budget > 0:
                     micro_batches = [[mb_rows[i] for i in b] for b in create_batches(
@@ -357,7 +376,8 @@ class RLHFTrainer:
                     is_last = mi == len(micro_batches) - 1
                     sync_ctx = self.reducer.no_sync() if not is_last else _nullctx()
                     with sync_ctx:
-                        loss, st = self._micro_step(td, micro, len(mb_rows))
+                        loss, st = self._micro_step(td, micro, len(mb_rows),
+                                                    zero_weight=zero_weight)
                     for k2, v in st.items():
                         if k2 in stats:
                             stats[k2].append(v)
@@ -382,7 +402,8 @@ class RLHFTrainer:
                 out[k] = sum(v) / len(v)
         return out
 
-    def _micro_step(self, td: TrainData, micro: list[int], mb_size: int):
+    def _micro_step(self, td: TrainData, micro: list[int], mb_size: int,
+                    zero_weight: bool = False):
         cfg = self.cfg
         mp = [td.prompts[i] for i in micro]
         mr = [td.responses[i] for i in micro]
@@ -413,10 +434,7 @@ class RLHFTrainer:
         loss, st = self.algo.loss(self, td, mb, new_logprobs, vpred)
         # scale: token-budget buckets re-scale by rows/minibatch rows
         # (grpo_r1_trainer.py:787-790); fixed micro-batches average over count.
-        if cfg.train_token_budget > 0:
-            scale = len(micro) / mb_size
-        else:
-            scale = len(micro) / mb_size
+        scale = 0.0 if zero_weight else len(micro) / mb_size
         (loss * scale).backward()
         with torch.no_grad():
             ratio = st.pop("ratio", None)

@@ -85,10 +85,49 @@ def _run_trainer_dp(rank, world, port, q, tmpdir):
             torch_dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("fn", [_run_reducer, _run_trainer_dp])
+def _run_sparse_grpo_dp(rank, world, port, q, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    try:
+        import torch
+        from nanorlhf_amd.algos import grpo
+        from nanorlhf_amd.algos.grpo import GRPOConfig
+        from nanorlhf_amd.data import hh_shaped_prompts
+        from nanorlhf_amd.models import CausalLM
+
+        cfg = GRPOConfig(model_preset="tiny", dtype="float32", use_lora=True,
+                         lora_r=4, lora_alpha=8, per_device_train_batch_size=2,
+                         gradient_accumulation_steps=1, num_mini_batches=2,
+                         total_episodes=8, sample_n=2, response_length=4,
+                         temperature=1.0, stop_token_id=1,
+                         output_dir=os.path.join(tmpdir, f"sp{rank}"),
+                         gradient_checkpointing=False, score_token_budget=256,
+                         sparse_filter=True, train_token_budget=64)
+        torch.manual_seed(0)
+        policy = CausalLM.from_preset("tiny")
+        ref = CausalLM.from_preset("tiny")
+        ref.load_state_dict(policy.state_dict())
+        prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=8)
+
+        def reward(seqs):
+            # rank 1 gets ALL-ZERO scores -> sparse filter drops everything
+            # there while rank 0 keeps rows: minibatch counts must still match
+            if int(os.environ["RANK"]) == 1:
+                return torch.zeros(len(seqs))
+            return torch.tensor([(sum(s) + len(s)) % 3 - 1.0 for s in seqs])
+
+        tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
+        tr.train(num_updates=1)
+        q.put((rank, True))
+    finally:
+        if torch_dist.is_initialized():
+            torch_dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("fn", [_run_reducer, _run_trainer_dp, _run_sparse_grpo_dp])
 def test_world2_gloo(fn, tmp_path):
     world = 2
-    port = 29531 if fn is _run_reducer else 29533
+    port = {id(_run_reducer): 29531, id(_run_trainer_dp): 29533}.get(id(fn), 29537)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     args = (world, port, q) if fn is _run_reducer else (world, port, q, str(tmp_path))
