@@ -12,8 +12,6 @@ from __future__ import annotations
 
 from dataclasses import dataclass
 
-import torch
-
 from ..config import RLHFConfig
 from . import functional as F
 from .trainer import AlgoSpec, Rollout, RLHFTrainer, TrainData

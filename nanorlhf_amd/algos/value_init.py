@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import torch
 
-from ..sampler.engine import SamplerEngine, SamplingParams
+from ..sampler.engine import SamplingParams
 from ..utils.seed import rank_seed
 from . import functional as F
 

@@ -8,7 +8,7 @@ examples/<algo>.py.  Batch-size algebra fields keep trl/reference names and
 semantics (grpo_trainer.py:216-247) so configs transfer."""
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass

@@ -9,7 +9,7 @@ directly (288 GB leaves room for a merged copy), and `unmerge()` drops it.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 import torch.nn as nn

@@ -14,7 +14,7 @@ Replaces the reference's transformers Qwen2 + flash_attention_2 stack
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch

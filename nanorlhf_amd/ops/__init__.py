@@ -11,9 +11,8 @@ Every hot op has two paths:
 from __future__ import annotations
 
 import importlib
-import os
 
-import torch
+import torch  # noqa: F401  (must load before _C: provides libc10)
 
 _EXT = None
 _EXT_ERR: Exception | None = None

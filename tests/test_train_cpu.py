@@ -125,3 +125,37 @@ def test_model_reward_path(tmp_path):
     prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=7)
     tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
     tr.train(num_updates=1)
+
+
+def test_step_every_microbatch_quirk_mode(tmp_path):
+    """Reference quirk flag (optimizer.step per micro-batch inside
+    accumulate(), grpo_trainer.py:692) must run end to end."""
+    cfg = _mk(ReinforceConfig, tmp_path, step_every_microbatch=True)
+    policy, ref = _models(11)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=11)
+    tr = reinforce.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_score_rows_matches_direct_forward(tmp_path):
+    """score_rows' bucketed packed scoring must equal a direct per-row
+    forward (logprob of each response token)."""
+    import torch.nn.functional as TF
+    from nanorlhf_amd.models import pack_sequences
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=1, score_token_budget=64)
+    policy, ref = _models(12)
+    prompts = hh_shaped_prompts(6, 1024, min_len=4, max_len=9, seed=12)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    rows_p = [p for p in prompts[:4]]
+    rows_r = [[5, 9, 2], [7, 1], [3, 3, 3, 3], [8]]
+    lp, ref_lp, ent, mask, _ = tr.score_rows(rows_p, rows_r, with_ref=True)
+    for i, (p, r) in enumerate(zip(rows_p, rows_r)):
+        ids, cu, mx, pos = pack_sequences([torch.tensor(list(p) + r)])
+        h = tr.policy(ids, tr.policy.train_ctx(cu, mx, pos))
+        logits = tr.policy.logits(h).float() / (cfg.temperature + 1e-7)
+        logp = TF.log_softmax(logits, dim=-1)
+        for t, tok in enumerate(r):
+            want = float(logp[len(p) - 1 + t, tok])
+            assert abs(float(lp[i, t]) - want) < 1e-3, (i, t)
+        assert mask[i, : len(r)].sum() == len(r)
+        assert mask[i, len(r):].sum() == 0
