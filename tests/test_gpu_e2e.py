@@ -82,3 +82,26 @@ def test_grpo_gpu_short_training():
     m = tr._last_metrics
     assert all(torch.isfinite(torch.tensor(float(v))) for k, v in m.items()
                if isinstance(v, (int, float))), m
+
+
+def test_offload_engine_roundtrip():
+    """Pinned-host offload → restore must preserve forward results
+    (the ref/reward-model shuttle path, utils/offload.py)."""
+    import torch
+    from nanorlhf_amd.models import CausalLM, pack_sequences
+    from nanorlhf_amd.utils.offload import OffloadEngine
+    m = _small_model(seed=3).eval()
+    ids, cu, mx, pos = pack_sequences([torch.randint(2, 4096, (24,))], device=DEV)
+    ctx = type(m).train_ctx(cu, mx, pos)
+    with torch.no_grad():
+        h0 = m(ids, ctx).clone()
+    eng = OffloadEngine(torch.device(DEV), enabled=True)
+    eng.model_to_host(m)
+    eng.synchronize()
+    assert all(p.device.type == "cpu" for p in m.parameters())
+    eng.model_to_device(m)
+    eng.synchronize()
+    assert all(p.device.type == "cuda" for p in m.parameters())
+    with torch.no_grad():
+        h1 = m(ids, ctx)
+    assert torch.equal(h0, h1)
