@@ -138,7 +138,8 @@ class RLHFTrainer:
 
         # sampler over the live policy
         pool_tokens = cfg.kv_pool_tokens or self._auto_pool_tokens(train_prompts)
-        self.sampler = SamplerEngine(self.policy, kv_pool_tokens=pool_tokens)
+        self.sampler = SamplerEngine(self.policy, kv_pool_tokens=pool_tokens,
+                                     kv_cache_dtype=getattr(cfg, "kv_cache_dtype", "bf16"))
 
         # optimizer over trainable params (+ value model for PPO joint update)
         params = [p for p in self.policy.parameters() if p.requires_grad]

@@ -77,6 +77,8 @@ class RLHFConfig:
     score_token_budget: int = 22 * 2316    # fwd-scoring bucket budget (ref :534)
     train_token_budget: int = 0            # 0 → derive from micro-batch rows
     kv_pool_tokens: int = 0                # 0 → auto from batch & lengths
+    kv_cache_dtype: str = "bf16"           # "fp8_e4m3": OCP fp8 paged KV (halves
+                                           # the decode KV stream; opt-in)
     offload_ref: bool | None = None        # None → auto by memory pressure
     offload_reward: bool | None = None
     offload_optimizer: bool = False

@@ -18,17 +18,58 @@
 #include "common.h"
 
 typedef __bf16 bf16x8k __attribute__((ext_vector_type(8)));
+typedef int i32x2k __attribute__((ext_vector_type(2)));
 
 DEVINL f32x4 mfma16k(bf16x8k a, bf16x8k b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// ---- OCP fp8 e4m3 KV support (gfx950-native cvt instructions) ----------
+DEVINL unsigned char f32_to_fp8(float x) {
+  return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(x, x, 0, false) & 0xff);
+}
+
+// load 8 cache elements starting at element index `idx` as a bf16 MFMA frag
+DEVINL bf16x8k load_frag_bf16(const short* base, long idx) {
+  s16x8 raw = *reinterpret_cast<const s16x8*>(base + idx);
+  return *reinterpret_cast<bf16x8k*>(&raw);
+}
+
+DEVINL bf16x8k load_frag_fp8(const unsigned char* base, long idx) {
+  i32x2k w = *reinterpret_cast<const i32x2k*>(base + idx);  // 8 bytes
+  float f[8];
+  f[0] = __builtin_amdgcn_cvt_f32_fp8(w[0], 0);
+  f[1] = __builtin_amdgcn_cvt_f32_fp8(w[0], 1);
+  f[2] = __builtin_amdgcn_cvt_f32_fp8(w[0], 2);
+  f[3] = __builtin_amdgcn_cvt_f32_fp8(w[0], 3);
+  f[4] = __builtin_amdgcn_cvt_f32_fp8(w[1], 0);
+  f[5] = __builtin_amdgcn_cvt_f32_fp8(w[1], 1);
+  f[6] = __builtin_amdgcn_cvt_f32_fp8(w[1], 2);
+  f[7] = __builtin_amdgcn_cvt_f32_fp8(w[1], 3);
+  s16x8 raw;
+#pragma unroll
+  for (int j = 0; j < 8; j++) raw[j] = f2bf(f[j]);
+  return *reinterpret_cast<bf16x8k*>(&raw);
+}
+
+template <typename CT>
+DEVINL bf16x8k load_kv_frag(const CT* base, long idx);
+template <>
+DEVINL bf16x8k load_kv_frag<short>(const short* base, long idx) {
+  return load_frag_bf16(base, idx);
+}
+template <>
+DEVINL bf16x8k load_kv_frag<unsigned char>(const unsigned char* base, long idx) {
+  return load_frag_fp8(base, idx);
+}
+
 // ------------------------------------------------------------- kv append
+template <typename CT>
 __global__ void kv_append_kernel(const short* __restrict__ k,
                                  const short* __restrict__ v,
                                  const long* __restrict__ slots,
-                                 short* __restrict__ kc,
-                                 short* __restrict__ vc,
+                                 CT* __restrict__ kc,
+                                 CT* __restrict__ vc,
                                  long T, int Hkv, int D, int ps) {
   const int row_elems = Hkv * D;
   long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;  // units of 8 elems
@@ -37,18 +78,27 @@ __global__ void kv_append_kernel(const short* __restrict__ k,
   const long t = idx / nvec;
   const int e0 = (int)(idx % nvec) * 8;
   const long slot = slots[t];
-  // K: token-major row copy
-  *reinterpret_cast<s16x8*>(kc + slot * (long)row_elems + e0) =
-      *reinterpret_cast<const s16x8*>(k + t * (long)row_elems + e0);
-  // V: d-major scatter — vc[((page*Hkv + h)*D + d)*ps + off]
+  s16x8 kk = *reinterpret_cast<const s16x8*>(k + t * (long)row_elems + e0);
+  s16x8 vv = *reinterpret_cast<const s16x8*>(v + t * (long)row_elems + e0);
   const long page = slot / ps;
   const int off = (int)(slot % ps);
-  s16x8 vv = *reinterpret_cast<const s16x8*>(v + t * (long)row_elems + e0);
   const int h = e0 / D;
   const int d0 = e0 % D;
+  if constexpr (sizeof(CT) == 2) {
+    // K: token-major row copy (bf16)
+    *reinterpret_cast<s16x8*>(kc + slot * (long)row_elems + e0) = kk;
 #pragma unroll
-  for (int j = 0; j < 8; j++)
-    vc[((page * Hkv + h) * (long)D + d0 + j) * ps + off] = vv[j];
+    for (int j = 0; j < 8; j++)
+      vc[((page * Hkv + h) * (long)D + d0 + j) * ps + off] = (CT)vv[j];
+  } else {
+    // fp8 e4m3 quantized cache
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      kc[slot * (long)row_elems + e0 + j] = (CT)f32_to_fp8(bf2f(kk[j]));
+      vc[((page * Hkv + h) * (long)D + d0 + j) * ps + off] =
+          (CT)f32_to_fp8(bf2f(vv[j]));
+    }
+  }
 }
 
 void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor slots,
@@ -62,18 +112,29 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor slots,
   TORCH_CHECK(D % 8 == 0);
   const long total = T * (Hkv * D / 8);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(kv_append_kernel, dim3((total + 255) / 256), dim3(256), 0, stream,
-                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     slots.data_ptr<long>(), (short*)k_cache.data_ptr(),
-                     (short*)v_cache.data_ptr(), T, Hkv, D, ps);
+  if (k_cache.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(kv_append_kernel<short>, dim3((total + 255) / 256), dim3(256),
+                       0, stream,
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       slots.data_ptr<long>(), (short*)k_cache.data_ptr(),
+                       (short*)v_cache.data_ptr(), T, Hkv, D, ps);
+  } else if (k_cache.scalar_type() == torch::kFloat8_e4m3fn) {
+    hipLaunchKernelGGL(kv_append_kernel<unsigned char>, dim3((total + 255) / 256),
+                       dim3(256), 0, stream,
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       slots.data_ptr<long>(), (unsigned char*)k_cache.data_ptr(),
+                       (unsigned char*)v_cache.data_ptr(), T, Hkv, D, ps);
+  } else {
+    TORCH_CHECK(false, "kv cache must be bf16 or fp8_e4m3fn");
+  }
   HIP_CHECK_LAST();
 }
 
 // ------------------------------------------------------- decode attention
-template <int D>
+template <int D, typename CT>
 __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D]
-                                    const short* __restrict__ kc,   // [P, 16, Hkv, D]
-                                    const short* __restrict__ vc,   // [P, Hkv, D, 16]
+                                    const CT* __restrict__ kc,      // [P, 16, Hkv, D]
+                                    const CT* __restrict__ vc,      // [P, Hkv, D, 16]
                                     const int* __restrict__ bt,     // [B, maxP]
                                     const int* __restrict__ seq_lens,
                                     short* __restrict__ out,        // [B, Hq, D]
@@ -126,8 +187,8 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
       const long krow = (((long)page * PS + lo) * Hkv + kvh) * D;
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        s16x8 raw = *reinterpret_cast<const s16x8*>(kc + krow + 32 * c + 8 * hi);
-        sc[n] = mfma16k(qf[c], *reinterpret_cast<bf16x8k*>(&raw), sc[n]);
+        bf16x8k kf = load_kv_frag<CT>(kc, krow + 32 * c + 8 * hi);
+        sc[n] = mfma16k(qf[c], kf, sc[n]);
       }
     }
     // ---- mask + online softmax (rows = q heads g = 4*hi + r) ----------
@@ -183,8 +244,8 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
 #pragma unroll
     for (int t = 0; t < D / 16; t++) {
       const long vaddr = (((long)pv_page * Hkv + kvh) * D + 16 * t + lo) * PS + tokoff;
-      s16x8 raw = *reinterpret_cast<const s16x8*>(vc + vaddr);
-      acc_o[t] = mfma16k(pf, *reinterpret_cast<bf16x8k*>(&raw), acc_o[t]);
+      bf16x8k vf = load_kv_frag<CT>(vc, vaddr);
+      acc_o[t] = mfma16k(pf, vf, acc_o[t]);
     }
   }
 
@@ -252,15 +313,18 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor k_cache,
   if (B == 0) return out;
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(B * Hkv), block(256);
-#define LAUNCH_D(DD)                                                            \
-  hipLaunchKernelGGL((paged_decode_kernel<DD>), grid, block, 0, stream,         \
-                     (const short*)q.data_ptr(), (const short*)k_cache.data_ptr(), \
-                     (const short*)v_cache.data_ptr(), block_tables.data_ptr<int>(), \
+  const bool fp8 = k_cache.scalar_type() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(fp8 || k_cache.scalar_type() == torch::kBFloat16,
+              "kv cache must be bf16 or fp8_e4m3fn");
+#define LAUNCH_D(DD, CT)                                                        \
+  hipLaunchKernelGGL((paged_decode_kernel<DD, CT>), grid, block, 0, stream,     \
+                     (const short*)q.data_ptr(), (const CT*)k_cache.data_ptr(), \
+                     (const CT*)v_cache.data_ptr(), block_tables.data_ptr<int>(), \
                      seq_lens.data_ptr<int>(), (short*)out.data_ptr(),          \
                      Hq, Hkv, maxP, G, (float)scale)
-  if (D == 128) { LAUNCH_D(128); }
-  else if (D == 64) { LAUNCH_D(64); }
-  else if (D == 32) { LAUNCH_D(32); }
+  if (D == 128) { if (fp8) LAUNCH_D(128, unsigned char); else LAUNCH_D(128, short); }
+  else if (D == 64) { if (fp8) LAUNCH_D(64, unsigned char); else LAUNCH_D(64, short); }
+  else if (D == 32) { if (fp8) LAUNCH_D(32, unsigned char); else LAUNCH_D(32, short); }
   else { TORCH_CHECK(false, "unsupported head_dim ", D); }
 #undef LAUNCH_D
   HIP_CHECK_LAST();

@@ -36,14 +36,22 @@ class SamplingParams:
 
 
 class SamplerEngine:
+    KV_DTYPES = {"bf16": torch.bfloat16, "fp8_e4m3": torch.float8_e4m3fn}
+
     def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
                  max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072,
-                 compact_interval: int = 16, use_graphs: bool = True):
+                 compact_interval: int = 16, use_graphs: bool = True,
+                 kv_cache_dtype: str = "bf16"):
         self.model = model
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
+        kv_dtype = (self.KV_DTYPES[kv_cache_dtype]
+                    if self.device.type == "cuda" or kv_cache_dtype != "bf16"
+                    else self.dtype)
+        if self.device.type != "cuda" and kv_cache_dtype == "bf16":
+            kv_dtype = self.dtype  # CPU tests may run fp32 models
         self.pool = PagedKVCache.for_budget(model.cfg, kv_pool_tokens, page_size,
-                                            device=self.device, dtype=self.dtype)
+                                            device=self.device, dtype=kv_dtype)
         self.max_num_seqs = max_num_seqs
         self.prefill_chunk_tokens = prefill_chunk_tokens
         self.compact_interval = compact_interval
