@@ -184,10 +184,25 @@ class RLHFTrainer:
 
     # ------------------------------------------------------------------ utils
     def _auto_pool_tokens(self, prompts) -> int:
+        from ..sampler.cache import PagedKVCache
+        from ..sampler.engine import SamplerEngine
         max_prompt = max((len(p) for p in prompts), default=256)
         per_seq = max_prompt + self.cfg.response_length
         n_seq = self.sizes["local_batch_size"] * max(self.cfg.sample_n, 1)
-        return per_seq * min(n_seq, 4096) + 4096
+        want = per_seq * min(n_seq, 4096) + 4096
+        if self.device.type == "cuda":
+            # cap the pool against free HBM (long-response configs like r1's
+            # 8000 tokens would otherwise over-allocate); continuous batching
+            # queues the sequences that don't fit concurrently
+            kv_dtype = SamplerEngine.KV_DTYPES.get(
+                getattr(self.cfg, "kv_cache_dtype", "bf16"), torch.bfloat16)
+            bpt = PagedKVCache.bytes_per_token(
+                self.policy.cfg, dtype=kv_dtype)
+            free, _ = torch.cuda.mem_get_info(self.device)
+            cap = int(free * 0.55) // bpt
+            if want > cap:
+                want = max(cap, per_seq * 8)  # keep at least a few sequences
+        return want
 
     def _next_prompts(self) -> list[list[int]]:
         n = self.sizes["local_batch_size"]

@@ -53,7 +53,6 @@ def finetune_value_model(trainer, num_prompts: int = 500, epochs: int = 8,
         total, count = 0.0, 0
         for s in range(0, len(rows), minibatch_rows):
             chunk = rows[s: s + minibatch_rows]
-            vp = []
             lens = []
             ids_list = []
             for i in chunk:

@@ -83,7 +83,7 @@ class DecodeBatch:
         slots = page * ps + pos % ps
         return self.last_tok, pos, slots, self.seq_lens, self.bt
 
-    def commit(self, tokens: torch.Tensor, pad_token_id: int):
+    def commit(self, tokens: torch.Tensor):
         """Record sampled tokens; advance lengths.  Fully IN-PLACE on the
         batch's state tensors so the whole step is hipGraph-capturable (no
         host sync, no tensor reassignment)."""

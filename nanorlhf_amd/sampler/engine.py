@@ -114,23 +114,21 @@ class SamplerEngine:
                     s.finished = True
 
     # --------------------------------------------------------------- decode
-    def _decode_step_eager(self, db: DecodeBatch, params: SamplingParams,
-                           pad_token_id: int):
+    def _decode_step_eager(self, db: DecodeBatch, params: SamplingParams):
         ids, pos, slots, seq_lens, bt = db.step_inputs()
         ctx = AttnContext(mode="decode", positions=pos,
                           kv_caches=self.pool.layers, slots=slots,
                           block_tables=bt, seq_lens=seq_lens)
         hidden = self.model(ids, ctx)
         tokens = self._sample_from_hidden(hidden, params)
-        db.commit(tokens, pad_token_id)
+        db.commit(tokens)
 
-    def _run_decode_step(self, db: DecodeBatch, params: SamplingParams,
-                         pad_token_id: int) -> int:
+    def _run_decode_step(self, db: DecodeBatch, params: SamplingParams) -> int:
         """One decode step; hipGraph-captured and replayed when the batch
         state is stable.  Returns the number of steps actually taken (capture
         does 2 warmup steps)."""
         if not self.use_graphs:
-            self._decode_step_eager(db, params, pad_token_id)
+            self._decode_step_eager(db, params)
             return 1
         if self._graph is None or self._graph_version != db.version:
             torch.cuda.synchronize()
@@ -138,12 +136,12 @@ class SamplerEngine:
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 for _ in range(2):
-                    self._decode_step_eager(db, params, pad_token_id)
+                    self._decode_step_eager(db, params)
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self._decode_step_eager(db, params, pad_token_id)
+                self._decode_step_eager(db, params)
             self._graph = g
             self._graph_version = db.version
             return 2  # the warmup steps ran; captured step executes on replay
@@ -205,7 +203,7 @@ class SamplerEngine:
                 # device-resident decode: no host sync inside the chunk
                 inner = 0
                 while len(db) and inner < self.compact_interval:
-                    inner += self._run_decode_step(db, params, pad_token_id)
+                    inner += self._run_decode_step(db, params)
                 done.extend(db.compact())
                 outer_guard += 1
                 if outer_guard > 8 * (params.max_tokens // self.compact_interval + 2) \

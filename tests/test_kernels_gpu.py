@@ -184,7 +184,6 @@ def test_kv_append_and_paged_decode():
     q = _mt(B, Hq, D, seed=7)
     sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
     out = ops.paged_attn_decode(q, kc, vc, tables.to(DEV), sl, scale=D ** -0.5)
-    ref = ops.kvcache.paged_attn_decode.__wrapped__ if False else None
     # CPU reference via the ops CPU path
     out_ref = ops.paged_attn_decode(q.cpu(), kc.cpu(), vc.cpu(), tables, sl.cpu(),
                                     scale=D ** -0.5)
