@@ -36,6 +36,8 @@ def main():
     ap.add_argument("--response-length", type=int, default=1500)
     ap.add_argument("--prompts-per-rank", type=int, default=512)
     ap.add_argument("--sample-n", type=int, default=4)
+    ap.add_argument("--kv-dtype", type=str, default="bf16",
+                    help="bf16 (headline) | fp8_e4m3 (secondary measurement)")
     args = ap.parse_args()
 
     from nanorlhf_amd.algos import grpo
@@ -68,6 +70,7 @@ def main():
         gradient_checkpointing=False,
         score_token_budget=98304,
         train_token_budget=49152,
+        kv_cache_dtype=args.kv_dtype,
         output_dir=os.environ.get("BENCH_OUT", "/tmp/nanorlhf_bench"),
         save_steps=0, log_samples=0, report_to="none",
         missing_eos_penalty=1.0,
@@ -136,6 +139,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": eps_per_sec / 1.0,  # reference: ~1 episode/sec (1 s/episode, A100-40G)
             "dtype": "bf16",
+            "kv_cache_dtype": args.kv_dtype,
             "data": "synthetic hh-rlhf-shaped prompts, random-init weights (no network)",
             "config": {
                 "model": "qwen2.5-1.5b-instruct-arch (random init)",

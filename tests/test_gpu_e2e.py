@@ -105,3 +105,26 @@ def test_offload_engine_roundtrip():
     with torch.no_grad():
         h1 = m(ids, ctx)
     assert torch.equal(h0, h1)
+
+
+def test_7b_geometry_small_depth():
+    """Qwen2.5-7B geometry (28 q heads / 4 kv heads -> GQA G=7, untied
+    lm_head) through rollout + fwd/bwd at reduced depth."""
+    import torch
+    from nanorlhf_amd import ops
+    from nanorlhf_amd.models import CausalLM, get_config, pack_sequences
+    from nanorlhf_amd.sampler import SamplerEngine, SamplingParams
+    torch.manual_seed(0)
+    cfg = get_config("qwen2.5-7b", num_layers=2, vocab_size=8192)
+    m = CausalLM(cfg).to(DEV).to(torch.bfloat16)
+    eng = SamplerEngine(m, kv_pool_tokens=16384)
+    prompts = [torch.randint(2, 8192, (n,)).tolist() for n in (7, 30)]
+    out = eng.generate(prompts, SamplingParams(n=2, temperature=0.8, top_p=0.95,
+                                               max_tokens=8, seed=3))
+    assert out.shape == (4, 8) and (out >= 0).all()
+    ids, cu, mx, pos = pack_sequences([torch.randint(2, 8192, (40,))], device=DEV)
+    h = m(ids, CausalLM.train_ctx(cu, mx, pos))
+    lp, _ = ops.token_logprob_entropy(h, m.lm_head_weight, torch.roll(ids, -1), 1.0)
+    (-lp.mean()).backward()
+    g = m.model.layers[0].self_attn.qkv_proj.weight.grad
+    assert g is not None and torch.isfinite(g.float()).all()
