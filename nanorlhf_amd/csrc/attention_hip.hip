@@ -105,25 +105,45 @@ __global__ void fa_fwd_kernel(const short* __restrict__ q,
   for (int t = 0; t < D / 16; t++) acc_o[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = CAUSAL ? min(len, q0 + QTILE) : len;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KTILE) {
-    // ---- cooperative K/V stage (Kt row-major, Vt transposed) ----------
-    __syncthreads();
-    for (int idx = threadIdx.x; idx < KTILE * D / 8; idx += 256) {
+  // async-STAGE split (cdna_hip_programming.md G15): each thread owns
+  // NSLICE 16-B pieces of the KV tile; the NEXT tile's global loads are
+  // issued before computing on the CURRENT tile so HBM latency hides under
+  // the MFMA phase.  Loads are unconditional with a clamped token index
+  // (branching per element would force a vmcnt(0) drain per load — guide
+  // §5.4 trap 4(c)); out-of-range tokens are neutralized by the score mask
+  // (P = 0), so garbage rows never contribute.
+  constexpr int NSLICE = KTILE * D / 8 / 256;
+  s16x8 pk_[NSLICE], pv_[NSLICE];
+  auto issue_tile_loads = [&](int kv0) {
+#pragma unroll
+    for (int sl = 0; sl < NSLICE; sl++) {
+      const int idx = threadIdx.x + sl * 256;
       const int tok = idx / (D / 8);
       const int d0 = (idx % (D / 8)) * 8;
-      const int kvi = kv0 + tok;
-      s16x8 kk{}, vv{};
-      if (kvi < len) {
-        const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
-        kk = *reinterpret_cast<const s16x8*>(k + b);
-        vv = *reinterpret_cast<const s16x8*>(v + b);
-      }
-      *reinterpret_cast<s16x8*>(&Kt[swz_idx<2 * D>(tok, d0 * 2)]) = kk;
+      const int kvi = min(kv0 + tok, len - 1);
+      const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
+      pk_[sl] = *reinterpret_cast<const s16x8*>(k + b);
+      pv_[sl] = *reinterpret_cast<const s16x8*>(v + b);
+    }
+  };
+  auto write_tile_lds = [&]() {
+#pragma unroll
+    for (int sl = 0; sl < NSLICE; sl++) {
+      const int idx = threadIdx.x + sl * 256;
+      const int tok = idx / (D / 8);
+      const int d0 = (idx % (D / 8)) * 8;
+      *reinterpret_cast<s16x8*>(&Kt[swz_idx<2 * D>(tok, d0 * 2)]) = pk_[sl];
 #pragma unroll
       for (int j = 0; j < 8; j++)
-        Vt[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = vv[j];
+        Vt[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = pv_[sl][j];
     }
+  };
+  if (kv_end > 0) issue_tile_loads(0);
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KTILE) {
     __syncthreads();
+    write_tile_lds();
+    __syncthreads();
+    if (kv0 + KTILE < kv_end) issue_tile_loads(kv0 + KTILE);
     if (!wave_active) continue;
 
     // ---- S = Q K^T : NSUB 16-col subtiles ------------------------------
