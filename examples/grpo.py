@@ -65,7 +65,11 @@ if __name__ == "__main__":
     rm = ScalarHeadModel.from_preset(config.reward_preset)
     if ON_GPU:
         rm = rm.to(device).to(torch.bfloat16)
-    reward_fn = ModelReward(rm, device)
+    # host-offload policy for the RM (the reference shuttles it CPU<->GPU per
+    # reward pass, grpo.py:164,195; on 288 GB the auto policy keeps it
+    # resident and only offloads under real pressure)
+    from nanorlhf_amd.utils.offload import OffloadEngine
+    reward_fn = ModelReward(rm, device, offload=OffloadEngine(device) if ON_GPU else None)
 
     # synthetic hh-rlhf-shaped prompts (grpo.py:249-270 prompt prep)
     prompts = hh_shaped_prompts(2048, mcfg.vocab_size, seed=0)

@@ -284,3 +284,22 @@ def test_fp8_kv_sampler_end_to_end():
     # fp8 KV changes rounding; with near-flat random-init logits allow token
     # drift but the FIRST token (pure prefill, bf16 path) must match
     assert torch.equal(out_bf16[:, 0], out_fp8[:, 0])
+
+
+def test_ce_and_sampler_odd_vocab_tail():
+    """V not divisible by 8 exercises the scalar tail paths."""
+    V, N, H = 1003, 9, 64
+    hidden = _mt(N, H, scale=0.5, seed=31)
+    weight = _mt(V, H, scale=0.05, seed=32)
+    labels = torch.randint(0, V, (N,), device=DEV)
+    lp, ent = ops.token_logprob_entropy(hidden, weight, labels, temperature=1.0)
+    logits = (hidden.detach().cpu().float() @ weight.detach().cpu().float().t())
+    lse = torch.logsumexp(logits, -1)
+    lpr = logits.gather(-1, labels.cpu().unsqueeze(1)).squeeze(1) - lse
+    assert rel_err(lp.cpu(), lpr) < 3e-2
+    # greedy sampling over the odd vocab
+    lg = _mt(5, V, seed=33)
+    t = ops.sample_tokens(lg, 0.0, 1.0, seed=0, step=0)
+    assert torch.equal(t.cpu(), lg.float().argmax(-1).cpu())
+    t2 = ops.sample_tokens(lg, 1.0, 0.9, seed=4, step=2)
+    assert ((t2 >= 0) & (t2 < V)).all()

@@ -74,3 +74,15 @@ def test_sampling_temperature_variability():
     params2 = SamplingParams(n=2, temperature=1.5, top_p=0.95, max_tokens=6, seed=8)
     out3 = eng.generate(p, params2)
     assert not torch.equal(out1, out3)  # reseed changes rollouts (ref :127)
+
+
+def test_max_num_seqs_cap():
+    """Engine must respect the concurrent-sequence cap and still finish."""
+    torch.manual_seed(0)
+    m = CausalLM.from_preset("tiny").eval()
+    eng = SamplerEngine(m, kv_pool_tokens=8192, page_size=16, max_num_seqs=2)
+    prompts = [torch.randint(2, 1000, (8,)).tolist() for _ in range(5)]
+    params = SamplingParams(n=1, temperature=0.0, top_p=1.0, max_tokens=4, seed=1)
+    out = eng.generate(prompts, params)
+    big = SamplerEngine(m, kv_pool_tokens=8192, page_size=16)
+    assert torch.equal(out, big.generate(prompts, params))
