@@ -119,6 +119,17 @@ def call_with_timeout(fn, args, timeout_s: float) -> bool:
         return False
 
 
+def extract_math_answer(text: str) -> str | None:
+    """Boxed answer if present, else the last number in the text (the
+    reference's answer-extraction strategies, utils/data_processing/
+    answer_extraction.py:177-245 role)."""
+    pred = extract_boxed(text)
+    if pred is not None:
+        return pred
+    nums = re.findall(r"-?\d+(?:\.\d+)?", text)
+    return nums[-1] if nums else None
+
+
 def answers_equal(pred: str, gold: str, sympy_timeout_s: float = 0.5) -> bool:
     """3-stage equivalence (latex_answer_check.py:166-236 role)."""
     a, b = normalize_answer(pred), normalize_answer(gold)
@@ -134,3 +145,22 @@ def answers_equal(pred: str, gold: str, sympy_timeout_s: float = 0.5) -> bool:
     import sympy  # noqa: F401
     import sympy.parsing.sympy_parser  # noqa: F401
     return call_with_timeout(_sympy_worker, (a, b), sympy_timeout_s)
+
+
+# API-parity aliases for the reference's verifier entry points
+# (latex_answer_check.py:166 latex_answer_check; eval_script.py:6 is_correct;
+# eval_utils.py:181 math_equal; metamath_utils.py:171 is_equiv)
+def latex_answer_check(pred: str, gold: str, timeout_s: float = 0.5) -> bool:
+    return answers_equal(pred, gold, timeout_s)
+
+
+def is_correct(pred: str, gold: str, timeout_s: float = 0.5) -> bool:
+    return answers_equal(pred, gold, timeout_s)
+
+
+def math_equal(pred: str, gold: str, timeout_s: float = 0.5) -> bool:
+    return answers_equal(pred, gold, timeout_s)
+
+
+def is_equiv(pred: str, gold: str, timeout_s: float = 0.5) -> bool:
+    return answers_equal(pred, gold, timeout_s)

@@ -143,3 +143,15 @@ def test_disable_dropout():
     m = torch.nn.Sequential(torch.nn.Linear(2, 2), torch.nn.Dropout(0.5))
     disable_dropout_in_model(m)
     assert m[1].p == 0.0
+
+
+def test_math_aliases_and_extraction():
+    from nanorlhf_amd.rewards.mathcheck import (extract_math_answer, is_correct,
+                                                is_equiv, latex_answer_check,
+                                                math_equal)
+    assert extract_math_answer(r"thus \boxed{12}") == "12"
+    assert extract_math_answer("the result is 3.5 meters") == "3.5"
+    assert extract_math_answer("no numbers") is None
+    for fn in (is_correct, is_equiv, latex_answer_check, math_equal):
+        assert fn("42", "42.0")
+        assert not fn("41", "42")
