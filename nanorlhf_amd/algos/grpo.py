@@ -49,7 +49,9 @@ class GRPO(AlgoSpec):
             rows = keep
         prompts = [ro.prompts[i] for i in rows]
         responses = [ro.responses[i] for i in rows]
-        lp, ref_lp, ent, mask, _ = trainer.score_rows(prompts, responses, with_ref=True)
+        lp, ref_lp, ent, mask, _ = trainer.score_rows(
+            prompts, responses, with_ref=True,
+            rollout_lp=trainer.rollout_lp_for(ro, rows))
         adv_seq_kept = adv_seq[rows].to(trainer.device)
         eos_idx = mask.sum(1).long() - 1
         rewards = F.sparse_reward_at_eos(adv_seq_kept, mask, eos_idx)

@@ -50,6 +50,7 @@ class Rollout:
     raw_scores: torch.Tensor          # [R] before penalties
     contains_eos: torch.Tensor        # [R] bool
     sample_n: int
+    logprobs: Optional[list] = None   # sampler-reported per-token logprobs
 
     @property
     def num_rows(self):
@@ -241,7 +242,8 @@ class RLHFTrainer:
                                 top_p=cfg.top_p, max_tokens=cfg.response_length,
                                 seed=seed, stop_token_id=cfg.stop_token_id)
         with self.timers.phase("rollout"):
-            resp_pad = self.sampler.generate(prompts, params, pad_token_id=cfg.pad_token_id)
+            resp_pad, lp_pad = self.sampler.generate(
+                prompts, params, pad_token_id=cfg.pad_token_id, return_logprobs=True)
         greedy_scores = None
         if self.algo.greedy_baseline:
             gparams = SamplingParams(n=1, temperature=0.0, top_p=1.0,
@@ -250,6 +252,7 @@ class RLHFTrainer:
             with self.timers.phase("rollout_greedy"):
                 greedy_pad = self.sampler.generate(prompts, gparams,
                                                    pad_token_id=cfg.pad_token_id)
+
 
         def depad(resp_rows, n):
             rows_p, rows_r = [], []
@@ -265,10 +268,12 @@ class RLHFTrainer:
             return rows_p, rows_r
 
         rows_p, rows_r = depad(resp_pad, cfg.sample_n)
+        rows_lp = [lp_pad[i, : len(r)].tolist() for i, r in enumerate(rows_r)]
         with self.timers.phase("reward"):
             raw_scores = self.reward_fn([p + r for p, r in zip(rows_p, rows_r)]).float()
             if self.algo.greedy_baseline:
-                gp, gr = depad(greedy_pad, 1)
+                gp, gr = depad(greedy_pad if not isinstance(greedy_pad, tuple)
+                               else greedy_pad[0], 1)
                 greedy_scores = self.reward_fn([p + r for p, r in zip(gp, gr)]).float()
         contains_eos = torch.tensor(
             [cfg.stop_token_id is not None and (cfg.stop_token_id in r) for r in rows_r])
@@ -276,7 +281,8 @@ class RLHFTrainer:
         if cfg.missing_eos_penalty is not None and cfg.stop_token_id is not None:
             scores = torch.where(contains_eos, scores, scores - cfg.missing_eos_penalty)
         ro = Rollout(prompts=rows_p, responses=rows_r, scores=scores,
-                     raw_scores=raw_scores, contains_eos=contains_eos, sample_n=cfg.sample_n)
+                     raw_scores=raw_scores, contains_eos=contains_eos,
+                     sample_n=cfg.sample_n, logprobs=rows_lp)
         return ro, greedy_scores
 
     # --------------------------------------------------------------- scoring
@@ -304,9 +310,15 @@ class RLHFTrainer:
                 torch.tensor(row_of, dtype=torch.long, device=device),
                 torch.tensor(col_of, dtype=torch.long, device=device))
 
+    def rollout_lp_for(self, ro: Rollout, rows: list[int]):
+        """Per-row sampler logprobs when cfg.use_rollout_logprobs, else None."""
+        if not getattr(self.cfg, "use_rollout_logprobs", False) or ro.logprobs is None:
+            return None
+        return [ro.logprobs[i] for i in rows]
+
     @torch.no_grad()
     def score_rows(self, prompts, responses, with_ref: bool = True,
-                   with_values: bool = False):
+                   with_values: bool = False, rollout_lp=None):
         """Policy (+ref) logprobs and entropy for response tokens, padded
         [R, Lmax]; the reference's chunked scoring pass (grpo_trainer.py:534-577)
         with token-budget buckets (r1's _create_batches) instead of fixed rows."""
@@ -323,17 +335,25 @@ class RLHFTrainer:
         if self.ref_policy is not None and with_ref:
             self.offload.model_to_device(self.ref_policy)
             self.offload.synchronize()
+        if rollout_lp is not None:
+            # behavior-policy logprobs came from the sampler itself
+            for i, lps in enumerate(rollout_lp):
+                n = len(responses[i])
+                if n:
+                    logprobs[i, :n] = torch.tensor(lps[:n], device=device)
+                    mask[i, :n] = 1.0
         for bucket in create_batches(lens, cfg.score_token_budget):
             bp = [prompts[i] for i in bucket]
             br = [responses[i] for i in bucket]
             ids, cu, mx, pos, fidx, flab, frow, fcol = self._pack(bp, br)
-            ctx = AttnContext(mode="train", positions=pos, cu_seqlens=cu, max_seqlen=mx)
-            hidden = self.policy(ids, ctx)
-            lp, ent = ops.token_logprob_entropy(hidden[fidx], self.policy.lm_head_weight,
-                                                flab, cfg.temperature)
             rows = torch.tensor(bucket, device=device)[frow.cpu()].to(device)
-            logprobs[rows, fcol] = lp.to(logprobs.dtype)
-            entropy[rows, fcol] = ent.to(entropy.dtype)
+            if rollout_lp is None:
+                ctx = AttnContext(mode="train", positions=pos, cu_seqlens=cu, max_seqlen=mx)
+                hidden = self.policy(ids, ctx)
+                lp, ent = ops.token_logprob_entropy(hidden[fidx], self.policy.lm_head_weight,
+                                                    flab, cfg.temperature)
+                logprobs[rows, fcol] = lp.to(logprobs.dtype)
+                entropy[rows, fcol] = ent.to(entropy.dtype)
             mask[rows, fcol] = 1.0
             if with_ref and self.ref_policy is not None:
                 rctx = AttnContext(mode="train", positions=pos, cu_seqlens=cu, max_seqlen=mx)

@@ -43,6 +43,11 @@ class RLHFConfig:
     stop_token_id: int | None = None  # EOS; entry scripts set it
     pad_token_id: int = 0
     reseed_rollouts: bool = True      # new sampler seed per update (ref :127)
+    # use the sampler's own per-token logprobs as the behavior-policy
+    # ("old") logprobs, skipping the policy half of the scoring pass.  The
+    # reference recomputes them (its vLLM flow discards logprobs); default
+    # keeps that behavior, the flag is the in-process-sampler shortcut.
+    use_rollout_logprobs: bool = False
 
     # ---- objective ----------------------------------------------------------
     kl_coef: float = 0.05

@@ -19,8 +19,8 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor
 // sampling.hip
 torch::Tensor sample_topp(torch::Tensor logits, double temperature, double top_p,
                           long seed, long step);
-torch::Tensor sample_topp_dev(torch::Tensor logits, double temperature, double top_p,
-                              long seed, torch::Tensor step);
+std::vector<torch::Tensor> sample_topp_dev(torch::Tensor logits, double temperature,
+                                           double top_p, long seed, torch::Tensor step);
 // kvcache.hip
 void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor slots,
                torch::Tensor k_cache, torch::Tensor v_cache);

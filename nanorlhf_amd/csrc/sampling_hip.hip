@@ -28,6 +28,7 @@ DEVINL float warp_incl_scan(float v, int lane) {
 
 __global__ void sample_topp_kernel(const short* __restrict__ logits,
                                    long* __restrict__ out,
+                                   float* __restrict__ out_lp,  // nullable
                                    int V, float temperature, float top_p,
                                    unsigned long long seed,
                                    unsigned long long step_imm,
@@ -100,7 +101,11 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     __syncthreads();
   }
   if (temperature == 0.f) {
-    if (threadIdx.x == 0) out[row] = argmax_sh;
+    if (threadIdx.x == 0) {
+      out[row] = argmax_sh;
+      // greedy logprob reported under the T=1 softmax (vLLM convention)
+      if (out_lp) out_lp[row] = bf2f(lr[argmax_sh]) - (sm[0] + __logf(ss[0]));
+    }
     return;
   }
   const float M = sm[0];          // max of temperature-scaled logits
@@ -226,8 +231,11 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   }
   __syncthreads();
   if (threadIdx.x == 0) {
-    const int f = found_sh;
-    out[row] = (f >= 0) ? f : argmax_sh;  // rounding fallback: argmax
+    const int f = (found_sh >= 0) ? found_sh : argmax_sh;  // rounding fallback
+    out[row] = f;
+    // logprob of the chosen token under the temperature-scaled softmax
+    // (matches the scoring pass's logits/temperature quirk, grpo_trainer.py:547)
+    if (out_lp) out_lp[row] = bf2f(lr[f]) * invT - (M + __logf(ss[0]));
   }
 }
 
@@ -241,29 +249,32 @@ torch::Tensor sample_topp(torch::Tensor logits, double temperature, double top_p
   if (B == 0) return out;
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(sample_topp_kernel, dim3(B), dim3(SBLOCK), 0, stream,
-                     (const short*)logits.data_ptr(), out.data_ptr<long>(), V,
-                     (float)temperature, (float)top_p,
+                     (const short*)logits.data_ptr(), out.data_ptr<long>(), nullptr,
+                     V, (float)temperature, (float)top_p,
                      (unsigned long long)seed, (unsigned long long)step, nullptr);
   HIP_CHECK_LAST();
   return out;
 }
 
-torch::Tensor sample_topp_dev(torch::Tensor logits, double temperature, double top_p,
-                              long seed, torch::Tensor step) {
+std::vector<torch::Tensor> sample_topp_dev(torch::Tensor logits, double temperature,
+                                           double top_p, long seed, torch::Tensor step) {
   // step: int64 [1] device tensor, read inside the kernel — safe under
-  // hipGraph capture/replay (the immediate-arg form would freeze the step)
+  // hipGraph capture/replay (the immediate-arg form would freeze the step).
+  // Also returns the chosen token's logprob (vLLM logprobs parity).
   TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
   TORCH_CHECK(logits.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(step.scalar_type() == torch::kLong && step.is_cuda());
   const long B = logits.size(0);
   const int V = logits.size(1);
   auto out = torch::empty({B}, logits.options().dtype(torch::kLong));
-  if (B == 0) return out;
+  auto lp = torch::empty({B}, logits.options().dtype(torch::kFloat32));
+  if (B == 0) return {out, lp};
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(sample_topp_kernel, dim3(B), dim3(SBLOCK), 0, stream,
-                     (const short*)logits.data_ptr(), out.data_ptr<long>(), V,
+                     (const short*)logits.data_ptr(), out.data_ptr<long>(),
+                     lp.data_ptr<float>(), V,
                      (float)temperature, (float)top_p,
                      (unsigned long long)seed, 0ull, step.data_ptr<long>());
   HIP_CHECK_LAST();
-  return out;
+  return {out, lp};
 }

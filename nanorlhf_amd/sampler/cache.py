@@ -56,7 +56,8 @@ class PagedKVCache:
 class SeqState:
     """Book-keeping for one in-flight sequence."""
 
-    __slots__ = ("uid", "tokens", "prompt_len", "pages", "finished", "out_index")
+    __slots__ = ("uid", "tokens", "prompt_len", "pages", "finished", "out_index",
+                 "logprobs")
 
     def __init__(self, uid: int, prompt: list[int], out_index: int):
         self.uid = uid
@@ -65,6 +66,7 @@ class SeqState:
         self.pages: list[int] = []
         self.finished = False
         self.out_index = out_index
+        self.logprobs: list[float] = []  # per generated (response) token
 
     def __len__(self):
         return len(self.tokens)

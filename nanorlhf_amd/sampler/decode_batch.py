@@ -29,6 +29,7 @@ class DecodeBatch:
         self.last_tok = torch.zeros(0, dtype=torch.long, device=device)
         self.finished = torch.zeros(0, dtype=torch.bool, device=device)
         self.out = torch.zeros(0, max_tokens, dtype=torch.long, device=device)
+        self.lp = torch.zeros(0, max_tokens, dtype=torch.float32, device=device)
         self.version = 0  # bumped whenever state tensors are rebuilt
         if seqs:
             self.extend(seqs)
@@ -47,15 +48,17 @@ class DecodeBatch:
         last = torch.tensor([s.tokens[-1] for s in seqs], dtype=torch.long)
         fin = torch.tensor([s.finished for s in seqs], dtype=torch.bool)
         out = torch.zeros(B, self.max_tokens, dtype=torch.long)
+        lp = torch.zeros(B, self.max_tokens, dtype=torch.float32)
         gen = torch.zeros(B, dtype=torch.int32)
         for r, s in enumerate(seqs):
             # prefill already produced response token(s)
             resp = s.response
             if resp:
                 out[r, : len(resp)] = torch.tensor(resp, dtype=torch.long)
+                lp[r, : len(s.logprobs)] = torch.tensor(s.logprobs, dtype=torch.float32)
                 gen[r] = len(resp)
         return (bt.to(dev), seq_lens.to(dev), last.to(dev), fin.to(dev),
-                out.to(dev), gen.to(dev))
+                out.to(dev), gen.to(dev), lp.to(dev))
 
     def extend(self, seqs: list[SeqState]):
         pages = max([len(s.pages) for s in seqs] + [self.bt.shape[1]])
@@ -63,12 +66,13 @@ class DecodeBatch:
             pad = torch.zeros(len(self.seqs), pages - self.bt.shape[1],
                               dtype=torch.int32, device=self.device)
             self.bt = torch.cat([self.bt, pad], dim=1)
-        bt, sl, lt, fin, out, gen = self._rows_for(seqs, pages)
+        bt, sl, lt, fin, out, gen, lp = self._rows_for(seqs, pages)
         self.bt = torch.cat([self.bt, bt]) if len(self.seqs) else bt
         self.seq_lens = torch.cat([self.seq_lens, sl])
         self.last_tok = torch.cat([self.last_tok, lt])
         self.finished = torch.cat([self.finished, fin])
         self.out = torch.cat([self.out, out])
+        self.lp = torch.cat([self.lp, lp])
         self.gen_count = torch.cat([self.gen_count, gen])
         self.seqs.extend(seqs)
         self.version += 1
@@ -83,7 +87,7 @@ class DecodeBatch:
         slots = page * ps + pos % ps
         return self.last_tok, pos, slots, self.seq_lens, self.bt
 
-    def commit(self, tokens: torch.Tensor):
+    def commit(self, tokens: torch.Tensor, lps: torch.Tensor | None = None):
         """Record sampled tokens; advance lengths.  Fully IN-PLACE on the
         batch's state tensors so the whole step is hipGraph-capturable (no
         host sync, no tensor reassignment)."""
@@ -92,6 +96,10 @@ class DecodeBatch:
         cur = self.out.gather(1, idx).squeeze(1)
         val = torch.where(write, tokens, cur)
         self.out.scatter_(1, idx, val.unsqueeze(1))
+        if lps is not None:
+            cur_lp = self.lp.gather(1, idx).squeeze(1)
+            self.lp.scatter_(1, idx,
+                             torch.where(write, lps, cur_lp).unsqueeze(1))
         wi = write.int()
         self.gen_count.add_(wi)
         self.seq_lens.add_(wi)
@@ -110,6 +118,7 @@ class DecodeBatch:
         keep_mask = ~fin
         retired_rows = fin.nonzero(as_tuple=True)[0].tolist()
         out_cpu = self.out[fin.to(self.device)].cpu()
+        lp_cpu = self.lp[fin.to(self.device)].cpu()
         gen_cpu = self.gen_count[fin.to(self.device)].cpu()
         retired = []
         for j, r in enumerate(retired_rows):
@@ -119,6 +128,7 @@ class DecodeBatch:
             if self.stop is not None and self.stop in resp:
                 resp = resp[: resp.index(self.stop) + 1]
             s.tokens = s.tokens[: s.prompt_len] + resp
+            s.logprobs = lp_cpu[j, : len(resp)].tolist()
             s.finished = True
             self.pool.free(s.pages)
             s.pages = []
@@ -131,6 +141,7 @@ class DecodeBatch:
         self.last_tok = self.last_tok[km]
         self.finished = self.finished[km]
         self.out = self.out[km]
+        self.lp = self.lp[km]
         self.version += 1
         return retired
 

@@ -67,18 +67,26 @@ class SamplerEngine:
         return [seq.slot_of(p, ps) for p in range(start, end)]
 
     def _sample_from_hidden(self, hidden_last: torch.Tensor,
-                            params: SamplingParams) -> torch.Tensor:
+                            params: SamplingParams):
+        """Returns (tokens [B], logprobs [B] fp32) — the chosen token's
+        logprob under the temperature-scaled softmax (vLLM logprobs parity;
+        matches the scoring pass's logits/temperature quirk)."""
         logits = self.model.logits(hidden_last)
         if self._step_dev is not None:
             # device-side step counter: correct under hipGraph replay
             self._step_dev.add_(1)
-            return ops.ext().sample_topp_dev(logits.contiguous(),
-                                             float(params.temperature),
-                                             float(params.top_p),
-                                             int(params.seed), self._step_dev)
+            tok, lp = ops.ext().sample_topp_dev(logits.contiguous(),
+                                                float(params.temperature),
+                                                float(params.top_p),
+                                                int(params.seed), self._step_dev)
+            return tok, lp
         self._sample_step += 1
-        return ops.sample_tokens(logits, params.temperature, params.top_p,
-                                 params.seed, self._sample_step)
+        tok = ops.sample_tokens(logits, params.temperature, params.top_p,
+                                params.seed, self._sample_step)
+        inv_t = 1.0 / params.temperature if params.temperature > 0 else 1.0
+        logp = torch.log_softmax(logits.float() * inv_t, dim=-1)
+        lp = logp.gather(1, tok.unsqueeze(1)).squeeze(1)
+        return tok, lp
 
     # -------------------------------------------------------------- prefill
     @torch.no_grad()
@@ -106,10 +114,12 @@ class SamplerEngine:
                               max_seqlen=max(lens), kv_caches=self.pool.layers, slots=slots)
             hidden = self.model(ids, ctx)
             last_idx = cu[1:].long() - 1
-            tokens = self._sample_from_hidden(hidden[last_idx], params)
+            tokens, lps = self._sample_from_hidden(hidden[last_idx], params)
             tok_list = tokens.tolist()
-            for s, t in zip(chunk, tok_list):
+            lp_list = lps.tolist()
+            for s, t, l in zip(chunk, tok_list, lp_list):
                 s.tokens.append(int(t))
+                s.logprobs.append(float(l))
                 if params.stop_token_id is not None and int(t) == params.stop_token_id:
                     s.finished = True
 
@@ -120,8 +130,8 @@ class SamplerEngine:
                           kv_caches=self.pool.layers, slots=slots,
                           block_tables=bt, seq_lens=seq_lens)
         hidden = self.model(ids, ctx)
-        tokens = self._sample_from_hidden(hidden, params)
-        db.commit(tokens)
+        tokens, lps = self._sample_from_hidden(hidden, params)
+        db.commit(tokens, lps)
 
     def _run_decode_step(self, db: DecodeBatch, params: SamplingParams) -> int:
         """One decode step; hipGraph-captured and replayed when the batch
@@ -151,7 +161,8 @@ class SamplerEngine:
     # -------------------------------------------------------------- generate
     @torch.no_grad()
     def generate(self, prompts: list[list[int]], params: SamplingParams,
-                 pad_token_id: int = 0, merge_lora: bool = True) -> torch.Tensor:
+                 pad_token_id: int = 0, merge_lora: bool = True,
+                 return_logprobs: bool = False):
         """Sample params.n continuations per prompt.
 
         Returns LongTensor [len(prompts)*n, max_tokens]: responses
@@ -212,10 +223,15 @@ class SamplerEngine:
 
             out = torch.full((len(prompts) * params.n, params.max_tokens), pad_token_id,
                              dtype=torch.long)
+            lp_out = torch.zeros(len(prompts) * params.n, params.max_tokens)
             for s in done:
                 resp = s.response[: params.max_tokens]
                 if resp:
                     out[s.out_index, : len(resp)] = torch.tensor(resp, dtype=torch.long)
+                    lps = s.logprobs[: len(resp)]
+                    lp_out[s.out_index, : len(lps)] = torch.tensor(lps)
+            if return_logprobs:
+                return out, lp_out
             return out
         finally:
             if merge_lora:

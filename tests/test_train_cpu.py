@@ -159,3 +159,28 @@ def test_score_rows_matches_direct_forward(tmp_path):
             assert abs(float(lp[i, t]) - want) < 1e-3, (i, t)
         assert mask[i, : len(r)].sum() == len(r)
         assert mask[i, len(r):].sum() == 0
+
+
+def test_rollout_logprobs_match_scoring(tmp_path):
+    """The sampler's per-token logprobs (use_rollout_logprobs shortcut) must
+    agree with the recomputed scoring pass on an fp32 model."""
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2, use_lora=False,
+              temperature=0.9)
+    policy, ref = _models(21)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=21)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    ro, _ = tr._rollout(1)
+    assert ro.logprobs is not None and len(ro.logprobs) == ro.num_rows
+    lp, _, _, mask, _ = tr.score_rows(ro.prompts, ro.responses, with_ref=False)
+    for i in range(ro.num_rows):
+        n = len(ro.responses[i])
+        for t in range(n):
+            assert abs(ro.logprobs[i][t] - float(lp[i, t])) < 5e-3, (i, t)
+
+
+def test_grpo_with_rollout_logprobs(tmp_path):
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2, use_rollout_logprobs=True)
+    policy, ref = _models(22)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=22)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
