@@ -38,6 +38,9 @@ def main():
     ap.add_argument("--sample-n", type=int, default=4)
     ap.add_argument("--kv-dtype", type=str, default="bf16",
                     help="bf16 (headline) | fp8_e4m3 (secondary measurement)")
+    ap.add_argument("--rollout-logprobs", action="store_true",
+                    help="use sampler-reported logprobs (skips the policy half "
+                         "of the scoring pass; secondary measurement)")
     args = ap.parse_args()
 
     from nanorlhf_amd.algos import grpo
@@ -71,6 +74,7 @@ def main():
         score_token_budget=98304,
         train_token_budget=49152,
         kv_cache_dtype=args.kv_dtype,
+        use_rollout_logprobs=args.rollout_logprobs,
         output_dir=os.environ.get("BENCH_OUT", "/tmp/nanorlhf_bench"),
         save_steps=0, log_samples=0, report_to="none",
         missing_eos_penalty=1.0,
