@@ -52,7 +52,10 @@ __global__ void ce_rowstats_kernel(const short* __restrict__ logits,
   for (int off = 32; off > 0; off >>= 1) {
     float m2 = __shfl_xor(m, off), s2 = __shfl_xor(s, off), t2 = __shfl_xor(t, off);
     float mn = fmaxf(m, m2);
-    float c1 = __expf(m - mn), c2 = __expf(m2 - mn);
+    // -inf guard: threads with no elements (V < 8*BLOCK) would otherwise
+    // poison the merge with exp(-inf - -inf) = NaN
+    float c1 = (m == -INFINITY) ? 0.f : __expf(m - mn);
+    float c2 = (m2 == -INFINITY) ? 0.f : __expf(m2 - mn);
     s = s * c1 + s2 * c2;
     t = t * c1 + t2 * c2;
     m = mn;
@@ -63,7 +66,8 @@ __global__ void ce_rowstats_kernel(const short* __restrict__ logits,
     float M = sm[0], S = ss[0], T = st[0];
     for (int w = 1; w < BLOCK / 64; w++) {
       float mn = fmaxf(M, sm[w]);
-      float c1 = __expf(M - mn), c2 = __expf(sm[w] - mn);
+      float c1 = (M == -INFINITY) ? 0.f : __expf(M - mn);
+      float c2 = (sm[w] == -INFINITY) ? 0.f : __expf(sm[w] - mn);
       S = S * c1 + ss[w] * c2;
       T = T * c1 + st[w] * c2;
       M = mn;
