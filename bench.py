@@ -82,10 +82,14 @@ def main():
 
     torch.manual_seed(1234)
     mcfg = get_config("qwen2.5-1.5b")
+    # construct in bf16 directly: fp32 CPU init would transiently cost
+    # ~14 GB host RAM per rank (x8 ranks in the scaling run)
+    torch.set_default_dtype(torch.bfloat16)
     policy = CausalLM(mcfg)
     ref = CausalLM(mcfg)
     ref.load_state_dict(policy.state_dict())
     rm_model = ScalarHeadModel.from_preset("rm-large")
+    torch.set_default_dtype(torch.float32)
     prompts = hh_shaped_prompts(2048, mcfg.vocab_size, seed=7)
 
     # trainer wires dist/devices itself (reads RANK/LOCAL_RANK/WORLD_SIZE)
