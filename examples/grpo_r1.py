@@ -64,7 +64,7 @@ def gold_answer_for(prompt):
     return str(sum(prompt) % 1000)
 
 
-def make_eval_fn(eval_prompts, gold, reward):
+def make_eval_fn(eval_prompts, gold):
     def eval_fn(trainer):
         """Greedy MATH-500-style accuracy pass (grpo_r1.py:276-341)."""
         params = SamplingParams(n=1, temperature=0.0, top_p=1.0,
@@ -74,8 +74,7 @@ def make_eval_fn(eval_prompts, gold, reward):
                                        pad_token_id=trainer.cfg.pad_token_id)
         texts = [detokenize([t for t in row if t != trainer.cfg.pad_token_id])
                  for row in out.tolist()]
-        scores = reward(texts, keys=None) if not isinstance(gold, list) else \
-            MathRuleReward(gold, require_boxed=False)(texts)
+        scores = MathRuleReward(gold, require_boxed=False)(texts)
         lens = [(row != trainer.cfg.pad_token_id).sum() for row in out]
         return {"accuracy": float(scores.mean()),
                 "response_length": float(torch.tensor([float(l) for l in lens]).mean())}
@@ -95,7 +94,6 @@ if __name__ == "__main__":
 
     # rule reward over detokenized prompt+response text, gold looked up by
     # rollout order (trainer passes prompt+response id sequences)
-    base_reward = MathRuleReward(train_gold, require_boxed=False, timeout_s=0.5)
     prompt_index = {tuple(p): i for i, p in enumerate(train_prompts)}
 
     def reward_fn(sequences):
@@ -114,6 +112,6 @@ if __name__ == "__main__":
         return MathRuleReward(golds, require_boxed=False)(texts)
 
     trainer = grpo.make_trainer(config, policy, ref_policy, reward_fn, train_prompts)
-    trainer.eval_fn = make_eval_fn(eval_prompts, eval_gold, base_reward)
+    trainer.eval_fn = make_eval_fn(eval_prompts, eval_gold)
     trainer.train(num_updates=3)
     trainer.save()

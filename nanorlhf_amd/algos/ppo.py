@@ -34,7 +34,8 @@ class PPO(AlgoSpec):
         cfg = trainer.cfg
         rows = list(range(ro.num_rows))
         lp, ref_lp, ent, mask, values = trainer.score_rows(
-            ro.prompts, ro.responses, with_ref=True, with_values=True)
+            ro.prompts, ro.responses, with_ref=True, with_values=True,
+            rollout_lp=trainer.rollout_lp_for(ro, list(range(ro.num_rows))))
         values = values * mask
         eos_idx = mask.sum(1).long() - 1
         scores = ro.scores.to(trainer.device)

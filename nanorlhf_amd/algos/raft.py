@@ -32,7 +32,9 @@ class RAFT(AlgoSpec):
     def make_train_data(self, trainer: RLHFTrainer, ro: Rollout, greedy_scores=None) -> TrainData:
         cfg = trainer.cfg
         n = ro.sample_n
-        lp, ref_lp, ent, mask, _ = trainer.score_rows(ro.prompts, ro.responses, with_ref=True)
+        lp, ref_lp, ent, mask, _ = trainer.score_rows(
+            ro.prompts, ro.responses, with_ref=True,
+            rollout_lp=trainer.rollout_lp_for(ro, list(range(ro.num_rows))))
         kl_seq = ((lp - ref_lp) * mask).sum(1)
         rlhf_reward = ro.scores.to(trainer.device) - cfg.kl_coef * kl_seq
         g = rlhf_reward.view(-1, n)

@@ -26,7 +26,9 @@ class REINFORCE(AlgoSpec):
     def make_train_data(self, trainer: RLHFTrainer, ro: Rollout, greedy_scores=None) -> TrainData:
         cfg = trainer.cfg
         rows = list(range(ro.num_rows))
-        lp, ref_lp, ent, mask, _ = trainer.score_rows(ro.prompts, ro.responses, with_ref=True)
+        lp, ref_lp, ent, mask, _ = trainer.score_rows(
+            ro.prompts, ro.responses, with_ref=True,
+            rollout_lp=trainer.rollout_lp_for(ro, rows))
         eos_idx = mask.sum(1).long() - 1
         scores = ro.scores.to(trainer.device)
         rewards = F.kl_shaped_rewards(scores, lp, ref_lp, mask, eos_idx, cfg.kl_coef)

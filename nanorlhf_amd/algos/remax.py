@@ -28,7 +28,9 @@ class ReMax(AlgoSpec):
         cfg = trainer.cfg
         assert greedy_scores is not None
         rows = list(range(ro.num_rows))
-        lp, ref_lp, ent, mask, _ = trainer.score_rows(ro.prompts, ro.responses, with_ref=True)
+        lp, ref_lp, ent, mask, _ = trainer.score_rows(
+            ro.prompts, ro.responses, with_ref=True,
+            rollout_lp=trainer.rollout_lp_for(ro, rows))
         adv_scores = F.remax_advantage(ro.scores, greedy_scores).to(trainer.device)
         eos_idx = mask.sum(1).long() - 1
         rewards = F.kl_shaped_rewards(adv_scores, lp, ref_lp, mask, eos_idx, cfg.kl_coef)

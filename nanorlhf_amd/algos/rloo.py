@@ -31,7 +31,9 @@ class RLOO(AlgoSpec):
         cfg = trainer.cfg
         n = ro.sample_n
         # score ALL rows (KL enters the reward)
-        lp, ref_lp, ent, mask, _ = trainer.score_rows(ro.prompts, ro.responses, with_ref=True)
+        lp, ref_lp, ent, mask, _ = trainer.score_rows(
+            ro.prompts, ro.responses, with_ref=True,
+            rollout_lp=trainer.rollout_lp_for(ro, list(range(ro.num_rows))))
         kl_seq = ((lp - ref_lp) * mask).sum(1)            # [B*n]
         scores = ro.scores.to(trainer.device)
         rlhf_reward = scores - cfg.kl_coef * kl_seq       # sequence-level (:571-573)
