@@ -287,28 +287,37 @@ class RLHFTrainer:
 
     # --------------------------------------------------------------- scoring
     def _pack(self, prompts, responses):
+        """Pack rows into the varlen layout + response-token index maps.
+
+        Vectorized (torch ops on CPU, single H2D copies): the original
+        per-token Python list build cost ~1-2 s per full-scale update."""
         device = self.device
-        lens = [len(p) + len(r) for p, r in zip(prompts, responses)]
-        ids = torch.tensor([t for p, r in zip(prompts, responses) for t in (list(p) + list(r))],
-                           dtype=torch.long, device=device)
-        cu = torch.zeros(len(lens) + 1, dtype=torch.int32, device=device)
-        cu[1:] = torch.cumsum(torch.tensor(lens, dtype=torch.int32, device=device), 0)
-        pos = torch.cat([torch.arange(n, device=device) for n in lens])
-        # flat indices of hidden positions predicting each response token
-        flat_idx, flat_labels, row_of, col_of = [], [], [], []
-        off = 0
+        R = len(prompts)
+        pl = torch.tensor([len(p) for p in prompts], dtype=torch.long)
+        rl = torch.tensor([len(r) for r in responses], dtype=torch.long)
+        lens = pl + rl
+        cu_cpu = torch.zeros(R + 1, dtype=torch.long)
+        cu_cpu[1:] = torch.cumsum(lens, 0)
+        T = int(cu_cpu[-1])
+        ids_cpu = torch.empty(T, dtype=torch.long)
         for i, (p, r) in enumerate(zip(prompts, responses)):
-            pl, rl = len(p), len(r)
-            flat_idx.extend(range(off + pl - 1, off + pl - 1 + rl))
-            flat_labels.extend(r)
-            row_of.extend([i] * rl)
-            col_of.extend(range(rl))
-            off += pl + rl
-        return (ids, cu, max(lens), pos,
-                torch.tensor(flat_idx, dtype=torch.long, device=device),
-                torch.tensor(flat_labels, dtype=torch.long, device=device),
-                torch.tensor(row_of, dtype=torch.long, device=device),
-                torch.tensor(col_of, dtype=torch.long, device=device))
+            s0 = int(cu_cpu[i])
+            ids_cpu[s0: s0 + len(p)] = torch.as_tensor(p, dtype=torch.long)
+            ids_cpu[s0 + len(p): s0 + len(p) + len(r)] = torch.as_tensor(r, dtype=torch.long)
+        # positions 0..len-1 per row
+        pos_cpu = torch.arange(T) - torch.repeat_interleave(cu_cpu[:-1], lens)
+        # response-token maps: hidden index (pl-1+t within row), label, row, col
+        row_of_cpu = torch.repeat_interleave(torch.arange(R), rl)
+        rcu = torch.zeros(R + 1, dtype=torch.long)
+        rcu[1:] = torch.cumsum(rl, 0)
+        TR = int(rcu[-1])
+        col_of_cpu = torch.arange(TR) - torch.repeat_interleave(rcu[:-1], rl)
+        flat_idx_cpu = (cu_cpu[:-1] + pl - 1)[row_of_cpu] + col_of_cpu
+        flat_labels_cpu = ids_cpu[(cu_cpu[:-1] + pl)[row_of_cpu] + col_of_cpu]
+        return (ids_cpu.to(device), cu_cpu.to(torch.int32).to(device),
+                int(lens.max()) if R else 0, pos_cpu.to(device),
+                flat_idx_cpu.to(device), flat_labels_cpu.to(device),
+                row_of_cpu.to(device), col_of_cpu.to(device))
 
     def rollout_lp_for(self, ro: Rollout, rows: list[int]):
         """Per-row sampler logprobs when cfg.use_rollout_logprobs, else None."""
