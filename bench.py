@@ -71,7 +71,7 @@ def main():
         kl_coef=0.05, learning_rate=3e-6,
         # 288 GB HBM: no recompute needed at these batch shapes
         gradient_checkpointing=False,
-        score_token_budget=98304,
+        score_token_budget=131072,
         train_token_budget=49152,
         kv_cache_dtype=args.kv_dtype,
         use_rollout_logprobs=args.rollout_logprobs,

@@ -20,7 +20,7 @@ import torch
 
 from . import ext
 
-_DEF_CHUNK = 4096
+_DEF_CHUNK = 16384
 
 
 def _ref_rowstats(logits: torch.Tensor, labels: torch.Tensor):

@@ -40,7 +40,7 @@ class SamplerEngine:
 
     def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
                  max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072,
-                 compact_interval: int = 16, use_graphs: bool = True,
+                 compact_interval: int = 32, use_graphs: bool = True,
                  kv_cache_dtype: str = "bf16"):
         self.model = model
         self.device = next(model.parameters()).device
