@@ -184,3 +184,47 @@ def test_grpo_with_rollout_logprobs(tmp_path):
     prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=22)
     tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
     tr.train(num_updates=1)
+
+
+def test_metric_names_logged(tmp_path):
+    """The reference's stable metric names must appear in metrics.jsonl
+    (SURVEY §5 observability; dashboards transfer)."""
+    import json
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2)
+    policy, ref = _models(31)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=31)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+    rec = json.loads(open(os.path.join(tmp_path, "metrics.jsonl")).readline())
+    for name in ("objective/kl_old", "objective/entropy_old",
+                 "eval_objective/rlhf_reward_old", "policy/approxkl_avg_new",
+                 "policy/clipfrac_avg_new", "loss/policy_avg_new",
+                 "policy/entropy_avg_new", "val/ratio_new", "val/ratio_var_new",
+                 "val/num_eos_tokens_old", "lr", "episode"):
+        assert name in rec, name
+
+
+def test_multi_epoch_and_whiten_rewards(tmp_path):
+    """num_ppo_epochs=2 reuses the batch off-policy (clip active);
+    whiten_rewards normalizes the reward stream."""
+    cfg = _mk(GRPOConfig, tmp_path, sample_n=2, num_ppo_epochs=2,
+              whiten_rewards=True)
+    policy, ref = _models(32)
+    prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=10, seed=32)
+    tr = grpo.make_trainer(cfg, policy, ref, _varied_reward, prompts)
+    tr.train(num_updates=1)
+
+
+def test_lr_schedule_progression(tmp_path):
+    cfg = _mk(ReinforceConfig, tmp_path, total_episodes=16,
+              lr_scheduler_type="cosine_with_min_lr", min_lr_ratio=0.5,
+              learning_rate=1e-4)
+    policy, ref = _models(33)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=33)
+    tr = reinforce.make_trainer(cfg, policy, ref,
+                                lambda s: constant_reward(s), prompts)
+    lr0 = tr._lr()
+    tr.train(num_updates=2)
+    assert tr.lr_step > 0
+    assert tr._lr() < lr0  # cosine decays
+    assert tr._lr() >= cfg.learning_rate * cfg.min_lr_ratio - 1e-12
