@@ -355,7 +355,7 @@ class RLHFTrainer:
             bp = [prompts[i] for i in bucket]
             br = [responses[i] for i in bucket]
             ids, cu, mx, pos, fidx, flab, frow, fcol = self._pack(bp, br)
-            rows = torch.tensor(bucket, device=device)[frow.cpu()].to(device)
+            rows = torch.tensor(bucket, device=device)[frow]
             if rollout_lp is None:
                 ctx = AttnContext(mode="train", positions=pos, cu_seqlens=cu, max_seqlen=mx)
                 hidden = self.policy(ids, ctx)
