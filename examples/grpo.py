@@ -74,6 +74,12 @@ if __name__ == "__main__":
     # synthetic hh-rlhf-shaped prompts (grpo.py:249-270 prompt prep)
     prompts = hh_shaped_prompts(2048, mcfg.vocab_size, seed=0)
 
-    trainer = grpo.make_trainer(config, policy, ref_policy, reward_fn, prompts)
+    # early stopping wired like the reference (patience 10^6 — effectively
+    # disabled; grpo.py:89,281)
+    from nanorlhf_amd.utils.callbacks import EarlyStoppingCallback
+    trainer = grpo.make_trainer(
+        config, policy, ref_policy, reward_fn, prompts,
+        callbacks=[EarlyStoppingCallback(
+            metric="eval_objective/rlhf_reward_old", patience=10**6)])
     trainer.train(num_updates=3)
     trainer.save()

@@ -199,6 +199,12 @@ class SamplerEngine:
                     s = waiting[-1]
                     ps = self.pool.page_size
                     need = (len(s) + params.max_tokens + ps - 1) // ps
+                    if need > self.pool.num_pages:
+                        raise RuntimeError(
+                            f"sequence needs {need} pages (prompt {len(s)} + "
+                            f"max_tokens {params.max_tokens}) but the KV pool "
+                            f"has only {self.pool.num_pages}; raise "
+                            "kv_pool_tokens or lower response_length")
                     if need > self.pool.free_pages:
                         break
                     waiting.pop()

@@ -155,3 +155,13 @@ def test_math_aliases_and_extraction():
     for fn in (is_correct, is_equiv, latex_answer_check, math_equal):
         assert fn("42", "42.0")
         assert not fn("41", "42")
+
+
+def test_mathcheck_pathological_timeout_fast():
+    """sympy bombs must be cut off by the subprocess timeout quickly."""
+    import time
+    from nanorlhf_amd.rewards.mathcheck import answers_equal
+    t0 = time.time()
+    out = answers_equal("x**x**x**x**99999 + y", "z", 0.3)
+    assert out is False
+    assert time.time() - t0 < 5.0

@@ -228,3 +228,31 @@ def test_lr_schedule_progression(tmp_path):
     assert tr.lr_step > 0
     assert tr._lr() < lr0  # cosine decays
     assert tr._lr() >= cfg.learning_rate * cfg.min_lr_ratio - 1e-12
+
+
+def test_callback_stop_and_oversize_prompt_error(tmp_path):
+    from nanorlhf_amd.utils.callbacks import TrainerCallback
+
+    class StopNow(TrainerCallback):
+        def __init__(self):
+            self.calls = 0
+
+        def on_update_end(self, trainer, metrics):
+            self.calls += 1
+            return True
+
+    cfg = _mk(ReinforceConfig, tmp_path)
+    policy, ref = _models(41)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=8, seed=41)
+    cb = StopNow()
+    tr = reinforce.make_trainer(cfg, policy, ref,
+                                lambda s: constant_reward(s), prompts,
+                                callbacks=[cb])
+    tr.train(num_updates=5)
+    assert cb.calls == 1 and tr.global_step == 1
+
+    # a sequence that can never fit the pool must raise, not hang
+    from nanorlhf_amd.sampler import SamplerEngine, SamplingParams
+    eng = SamplerEngine(tr.policy, kv_pool_tokens=64, page_size=16)
+    with pytest.raises(RuntimeError, match="KV pool"):
+        eng.generate([[2] * 200], SamplingParams(n=1, max_tokens=50))
