@@ -46,7 +46,24 @@ def masked_var(values: torch.Tensor, mask: torch.Tensor, unbiased: bool = True) 
 
 
 def masked_whiten(values: torch.Tensor, mask: torch.Tensor, shift_mean: bool = True) -> torch.Tensor:
-    """Whiten `values` using masked moments (trl semantics, grpo_trainer.py:607,619)."""
+    """Whiten `values` using masked moments (trl semantics, grpo_trainer.py:607,619).
+
+    GPU: fused HIP masked-moments + apply kernels (no grad flows through the
+    whitening of advantages); CPU: plain torch reference."""
+    if values.is_cuda and not values.requires_grad and values.dtype == torch.float32:
+        from .. import ops as _ops
+        vc = values.contiguous()
+        mc = mask.contiguous().float()
+        (partials,) = _ops.ext().masked_moments(vc.view(-1), mc.view(-1))
+        sums = partials.sum(0)  # [sum, sumsq, count]
+        cnt = sums[2].clamp(min=1.0)
+        mean = sums[0] / cnt
+        var = (sums[1] / cnt - mean * mean) * cnt / (cnt - 1.0).clamp(min=1.0)
+        invstd = torch.rsqrt(var + 1e-8)
+        shift = mean if shift_mean else torch.zeros_like(mean)
+        out = _ops.ext().whiten_apply(vc.view(-1), float(mean), float(invstd),
+                                      float(shift))
+        return out.view_as(values)
     mean = masked_mean(values, mask)
     var = masked_var(values, mask)
     whitened = (values - mean) * torch.rsqrt(var + 1e-8)

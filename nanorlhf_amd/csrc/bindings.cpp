@@ -27,6 +27,9 @@ void kv_append(torch::Tensor k, torch::Tensor v, torch::Tensor slots,
 torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor k_cache,
                                 torch::Tensor v_cache, torch::Tensor block_tables,
                                 torch::Tensor seq_lens, double scale);
+// masked.hip
+std::vector<torch::Tensor> masked_moments(torch::Tensor v, torch::Tensor mask);
+torch::Tensor whiten_apply(torch::Tensor v, double mean, double invstd, double shift);
 // attention.hip
 std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor cu_seqlens,
@@ -50,6 +53,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_topp_dev", &sample_topp_dev);
   m.def("kv_append", &kv_append);
   m.def("paged_attn_decode", &paged_attn_decode);
+  m.def("masked_moments", &masked_moments);
+  m.def("whiten_apply", &whiten_apply);
   m.def("fa_fwd_varlen", &fa_fwd_varlen);
   m.def("fa_bwd_varlen", &fa_bwd_varlen);
 }

@@ -18,7 +18,7 @@ CSRC = os.path.join(ROOT, "nanorlhf_amd", "csrc")
 sources = [
     os.path.join(CSRC, f)
     for f in ["bindings.cpp", "elementwise.hip", "logprob.hip", "adamw.hip",
-              "sampling.hip", "kvcache.hip", "attention.hip"]
+              "sampling.hip", "kvcache.hip", "attention.hip", "masked.hip"]
 ]
 
 setup(

@@ -303,3 +303,14 @@ def test_ce_and_sampler_odd_vocab_tail():
     assert torch.equal(t.cpu(), lg.float().argmax(-1).cpu())
     t2 = ops.sample_tokens(lg, 1.0, 0.9, seed=4, step=2)
     assert ((t2 >= 0) & (t2 < V)).all()
+
+
+def test_masked_whiten_fused():
+    from nanorlhf_amd.algos import functional as Fn
+    torch.manual_seed(0)
+    v = (torch.randn(37, 53) * 2 + 0.7).to(DEV)
+    mask = (torch.rand(37, 53) > 0.4).float().to(DEV)
+    for shift in (True, False):
+        got = Fn.masked_whiten(v, mask, shift_mean=shift)
+        want = Fn.masked_whiten(v.cpu(), mask.cpu(), shift_mean=shift)
+        assert rel_err(got.cpu(), want) < 1e-3
