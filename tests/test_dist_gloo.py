@@ -11,10 +11,15 @@ import torch.distributed as torch_dist
 import torch.multiprocessing as mp
 
 
-def _run_reducer(rank, world, port, q):
-    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
-    torch_dist.init_process_group("gloo", rank=rank, world_size=world)
+def _init_pg(rank, world, store_path):
+    """File-store rendezvous: no TCP port collisions across repeated runs."""
+    torch_dist.init_process_group("gloo", rank=rank, world_size=world,
+                                  init_method=f"file://{store_path}")
+
+
+def _run_reducer(rank, world, store_path, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    _init_pg(rank, world, store_path)
     try:
         from nanorlhf_amd.parallel.ddp import GradReducer
         from nanorlhf_amd.parallel import dist as pdist
@@ -49,9 +54,9 @@ def _run_reducer(rank, world, port, q):
         torch_dist.destroy_process_group()
 
 
-def _run_trainer_dp(rank, world, port, q, tmpdir):
-    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+def _run_trainer_dp(rank, world, store_path, q, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    _init_pg(rank, world, store_path)
     try:
         import torch
         from nanorlhf_amd.algos import reinforce
@@ -85,9 +90,9 @@ def _run_trainer_dp(rank, world, port, q, tmpdir):
             torch_dist.destroy_process_group()
 
 
-def _run_sparse_grpo_dp(rank, world, port, q, tmpdir):
-    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+def _run_sparse_grpo_dp(rank, world, store_path, q, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    _init_pg(rank, world, store_path)
     try:
         import torch
         from nanorlhf_amd.algos import grpo
@@ -127,10 +132,11 @@ def _run_sparse_grpo_dp(rank, world, port, q, tmpdir):
 @pytest.mark.parametrize("fn", [_run_reducer, _run_trainer_dp, _run_sparse_grpo_dp])
 def test_world2_gloo(fn, tmp_path):
     world = 2
-    port = {id(_run_reducer): 29531, id(_run_trainer_dp): 29533}.get(id(fn), 29537)
+    store_path = str(tmp_path / "pg_store")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    args = (world, port, q) if fn is _run_reducer else (world, port, q, str(tmp_path))
+    args = (world, store_path, q) if fn is _run_reducer else \
+        (world, store_path, q, str(tmp_path))
     procs = [ctx.Process(target=fn, args=(r, *args)) for r in range(world)]
     for p in procs:
         p.start()
