@@ -60,14 +60,15 @@ def masked_whiten(values: torch.Tensor, mask: torch.Tensor, shift_mean: bool = T
         mean = sums[0] / cnt
         var = (sums[1] / cnt - mean * mean) * cnt / (cnt - 1.0).clamp(min=1.0)
         invstd = torch.rsqrt(var + 1e-8)
-        shift = mean if shift_mean else torch.zeros_like(mean)
+        # trl semantics: shift_mean=True -> zero-centered; False -> mean kept.
+        shift = torch.zeros_like(mean) if shift_mean else mean
         out = _ops.ext().whiten_apply(vc.view(-1), float(mean), float(invstd),
                                       float(shift))
         return out.view_as(values)
     mean = masked_mean(values, mask)
     var = masked_var(values, mask)
     whitened = (values - mean) * torch.rsqrt(var + 1e-8)
-    if shift_mean:
+    if not shift_mean:
         whitened = whitened + mean
     return whitened
 

@@ -56,7 +56,7 @@ class GRPO(AlgoSpec):
         eos_idx = mask.sum(1).long() - 1
         rewards = F.sparse_reward_at_eos(adv_seq_kept, mask, eos_idx)
         if cfg.whiten_rewards:
-            rewards = F.masked_whiten(rewards, mask, shift_mean=False) * mask
+            rewards = F.masked_whiten(rewards, mask, shift_mean=True) * mask
         adv_tok = F.reward_to_go(rewards, gamma=1.0) * mask
         if cfg.advantage_whiten:
             adv_tok = F.masked_whiten(adv_tok, mask) * mask

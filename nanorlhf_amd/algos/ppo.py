@@ -22,7 +22,7 @@ class PPOConfig(RLHFConfig):
     sample_n: int = 1
     value_learning_rate: float = 3e-6
     whiten_rewards: bool = False
-    whiten_advantages: bool = True   # trl whitens GAE advantages
+    whiten_advantages: bool = False  # reference default: PPO/ppo.py:166 advantage_whiten=False
 
 
 class PPO(AlgoSpec):
@@ -41,11 +41,11 @@ class PPO(AlgoSpec):
         scores = ro.scores.to(trainer.device)
         rewards = F.kl_shaped_rewards(scores, lp, ref_lp, mask, eos_idx, cfg.kl_coef)
         if cfg.whiten_rewards:
-            rewards = F.masked_whiten(rewards, mask, shift_mean=False) * mask
+            rewards = F.masked_whiten(rewards, mask, shift_mean=True) * mask
         adv, returns = F.gae(rewards, values, cfg.gamma, cfg.lam)
         adv = adv * mask
         returns = returns * mask
-        if getattr(cfg, "whiten_advantages", True):
+        if getattr(cfg, "whiten_advantages", False):
             adv = F.masked_whiten(adv, mask) * mask
         kl_old = F.masked_mean(lp - ref_lp, mask)
         return TrainData(rows=rows, prompts=ro.prompts, responses=ro.responses,

@@ -33,7 +33,7 @@ class REINFORCE(AlgoSpec):
         scores = ro.scores.to(trainer.device)
         rewards = F.kl_shaped_rewards(scores, lp, ref_lp, mask, eos_idx, cfg.kl_coef)
         if cfg.whiten_rewards:
-            rewards = F.masked_whiten(rewards, mask, shift_mean=False) * mask
+            rewards = F.masked_whiten(rewards, mask, shift_mean=True) * mask
         adv = F.reward_to_go(rewards, gamma=cfg.gamma) * mask
         if cfg.advantage_whiten:
             adv = F.masked_whiten(adv, mask) * mask

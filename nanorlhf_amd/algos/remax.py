@@ -35,7 +35,7 @@ class ReMax(AlgoSpec):
         eos_idx = mask.sum(1).long() - 1
         rewards = F.kl_shaped_rewards(adv_scores, lp, ref_lp, mask, eos_idx, cfg.kl_coef)
         if cfg.whiten_rewards:
-            rewards = F.masked_whiten(rewards, mask, shift_mean=False) * mask
+            rewards = F.masked_whiten(rewards, mask, shift_mean=True) * mask
         adv = F.reward_to_go(rewards, gamma=cfg.gamma) * mask
         if cfg.advantage_whiten:
             adv = F.masked_whiten(adv, mask) * mask

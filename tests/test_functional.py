@@ -12,10 +12,11 @@ def test_masked_mean_whiten():
     v = torch.tensor([[1.0, 2.0, 3.0, 99.0]])
     m = torch.tensor([[1.0, 1.0, 1.0, 0.0]])
     assert float(F.masked_mean(v, m)) == pytest.approx(2.0)
-    w = F.masked_whiten(v, m, shift_mean=False)
+    # trl semantics: shift_mean=True -> zero-centered, False -> mean preserved
+    w = F.masked_whiten(v, m, shift_mean=True)
     mean = float(F.masked_mean(w, m))
     assert mean == pytest.approx(0.0, abs=1e-4)
-    w2 = F.masked_whiten(v, m, shift_mean=True)
+    w2 = F.masked_whiten(v, m, shift_mean=False)
     assert float(F.masked_mean(w2, m)) == pytest.approx(2.0, abs=1e-4)
 
 

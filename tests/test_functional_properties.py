@@ -64,7 +64,7 @@ def test_masked_whiten_moments(b, t, seed):
     mask = (torch.rand(b, t, generator=g) > 0.3).float()
     if float(mask.sum()) < 2 or float(F.masked_var(v, mask)) < 1e-6:
         return
-    w = F.masked_whiten(v, mask, shift_mean=False)
+    w = F.masked_whiten(v, mask, shift_mean=True)
     assert abs(float(F.masked_mean(w, mask))) < 1e-3
     assert abs(float(F.masked_var(w, mask)) - 1.0) < 0.05
 
