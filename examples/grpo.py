@@ -53,11 +53,29 @@ config = GRPOConfig(
     report_to="none",                           # set "wandb" + WANDB_PROJECT
 )
 
+# Point this at a local HF Qwen2.5 checkpoint directory (config.json +
+# safetensors + tokenizer files) to train REAL weights on REAL text
+# (reference grpo.py:209-270).  None → synthetic demo (no network here).
+MODEL_PATH = None        # e.g. "/models/Qwen2.5-1.5B-Instruct"
+RM_PATH = None           # sequence-classifier reward checkpoint (optional)
+
 if __name__ == "__main__":
-    mcfg = get_config(config.model_preset)
-    policy = CausalLM(mcfg)
-    ref_policy = CausalLM(mcfg)
-    ref_policy.load_state_dict(policy.state_dict())
+    tokenizer = None
+    if MODEL_PATH is not None:
+        # real-model seam: HF safetensors -> fused layout + tokenizer
+        from nanorlhf_amd.data.tokenizer import load_tokenizer, prepare_hh_prompts
+        from nanorlhf_amd.models.hf_import import load_pretrained
+        policy = load_pretrained(MODEL_PATH)
+        ref_policy = load_pretrained(MODEL_PATH)
+        tokenizer = load_tokenizer(MODEL_PATH)
+        config.stop_token_id = tokenizer.eos_token_id
+        config.pad_token_id = tokenizer.pad_token_id
+        mcfg = policy.cfg
+    else:
+        mcfg = get_config(config.model_preset)
+        policy = CausalLM(mcfg)
+        ref_policy = CausalLM(mcfg)
+        ref_policy.load_state_dict(policy.state_dict())
 
     # model-based reward (grpo.py:162-198: deberta-v3-large RM shuttled
     # on/off GPU → here an OffloadEngine policy inside ModelReward)
@@ -71,8 +89,19 @@ if __name__ == "__main__":
     from nanorlhf_amd.utils.offload import OffloadEngine
     reward_fn = ModelReward(rm, device, offload=OffloadEngine(device) if ON_GPU else None)
 
-    # synthetic hh-rlhf-shaped prompts (grpo.py:249-270 prompt prep)
-    prompts = hh_shaped_prompts(2048, mcfg.vocab_size, seed=0)
+    if tokenizer is not None:
+        # real text path: hh-rlhf-style records through the chat scaffold
+        # (grpo.py:249-270).  Reads a local jsonl of {"chosen": transcript}
+        # records; reward over decoded strings via StringReward when a
+        # string reward_func is preferred over the id-level ModelReward.
+        import json as _json
+        DATA_JSONL = "data/hh_train.jsonl"
+        with open(DATA_JSONL) as f:
+            records = [_json.loads(line) for line in f]
+        prompts = prepare_hh_prompts(records, tokenizer, max_prompt_len=512)
+    else:
+        # synthetic hh-rlhf-shaped prompts (grpo.py:249-270 prompt prep)
+        prompts = hh_shaped_prompts(2048, mcfg.vocab_size, seed=0)
 
     # early stopping wired like the reference (patience 10^6 — effectively
     # disabled; grpo.py:89,281)

@@ -270,11 +270,11 @@ class RLHFTrainer:
         rows_p, rows_r = depad(resp_pad, cfg.sample_n)
         rows_lp = [lp_pad[i, : len(r)].tolist() for i, r in enumerate(rows_r)]
         with self.timers.phase("reward"):
-            raw_scores = self.reward_fn([p + r for p, r in zip(rows_p, rows_r)]).float()
+            raw_scores = self._call_reward(rows_p, rows_r)
             if self.algo.greedy_baseline:
                 gp, gr = depad(greedy_pad if not isinstance(greedy_pad, tuple)
                                else greedy_pad[0], 1)
-                greedy_scores = self.reward_fn([p + r for p, r in zip(gp, gr)]).float()
+                greedy_scores = self._call_reward(gp, gr)
         contains_eos = torch.tensor(
             [cfg.stop_token_id is not None and (cfg.stop_token_id in r) for r in rows_r])
         scores = raw_scores.clone()
@@ -284,6 +284,15 @@ class RLHFTrainer:
                      raw_scores=raw_scores, contains_eos=contains_eos,
                      sample_n=cfg.sample_n, logprobs=rows_lp)
         return ro, greedy_scores
+
+    def _call_reward(self, rows_p: list[list[int]], rows_r: list[list[int]]) -> torch.Tensor:
+        """Dispatch to the reward plug-in.  String-contract rewards
+        (rewards.StringReward) get the response ids too in r1 mode, matching
+        reference grpo_r1.py:250's (strings, responses_ids, tokenizer)."""
+        seqs = [p + r for p, r in zip(rows_p, rows_r)]
+        if getattr(self.reward_fn, "mode", None) == "r1":
+            return self.reward_fn(seqs, rows_r).float()
+        return self.reward_fn(seqs).float()
 
     # --------------------------------------------------------------- scoring
     def _pack(self, prompts, responses):
