@@ -87,7 +87,9 @@ if __name__ == "__main__":
     # reward pass, grpo.py:164,195; on 288 GB the auto policy keeps it
     # resident and only offloads under real pressure)
     from nanorlhf_amd.utils.offload import OffloadEngine
-    reward_fn = ModelReward(rm, device, offload=OffloadEngine(device) if ON_GPU else None)
+    reward_fn = ModelReward(
+        rm, device,
+        offload=OffloadEngine(device, enabled=config.offload_reward) if ON_GPU else None)
 
     if tokenizer is not None:
         # real text path: hh-rlhf-style records through the chat scaffold

@@ -352,7 +352,7 @@ class RLHFTrainer:
         lens = [len(p) + len(r) for p, r in zip(prompts, responses)]
         if self.ref_policy is not None and with_ref:
             self.offload.model_to_device(self.ref_policy)
-            self.offload.synchronize()
+            self.offload.join_compute()
         if rollout_lp is not None:
             # behavior-policy logprobs came from the sampler itself
             for i, lps in enumerate(rollout_lp):
@@ -518,9 +518,18 @@ class RLHFTrainer:
         for update in range(1, n_updates + 1):
             try:
                 t0 = time.time()
+                if cfg.offload_optimizer:
+                    # optimizer state → host for the rollout+scoring phases
+                    # (reference state_to_device, grpo_trainer.py:475)
+                    self.offload.optimizer_state_to(self.optimizer, "cpu")
                 ro, greedy_scores = self._rollout(update)
                 with self.timers.phase("score"):
                     td = self.algo.make_train_data(self, ro, greedy_scores)
+                if cfg.offload_optimizer:
+                    # back to HBM before the update (reference :625); the
+                    # compute stream waits on the async copies, host doesn't
+                    self.offload.optimizer_state_to(self.optimizer, self.device)
+                    self.offload.join_compute()
                 with self.timers.phase("update"):
                     upd_stats = self._update(td)
             except Exception:

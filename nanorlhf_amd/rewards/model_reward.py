@@ -33,7 +33,7 @@ class ModelReward:
         string path lives in the entry scripts when a tokenizer is given.)"""
         if self.offload is not None:
             self.offload.model_to_device(self.model)
-            self.offload.synchronize()
+            self.offload.join_compute()
         self.model.eval()
         if self.max_len:
             sequences = [s[: self.max_len] for s in sequences]

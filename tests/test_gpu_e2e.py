@@ -107,6 +107,97 @@ def test_offload_engine_roundtrip():
     assert torch.equal(h0, h1)
 
 
+def test_offload_orders_after_pending_compute():
+    """model_to_host must see the RESULT of kernels still in flight on the
+    compute stream (round-1 race: the side-stream copy launched without
+    waiting).  Queue a long dependent-chain write into the params, then
+    offload immediately with no manual sync."""
+    from nanorlhf_amd.utils.offload import OffloadEngine
+    m = torch.nn.Linear(4096, 4096, bias=False).to(DEV)
+    eng = OffloadEngine(torch.device(DEV), enabled=True)
+    with torch.no_grad():
+        # long chain on the default stream that finally overwrites the weight
+        x = torch.randn(4096, 4096, device=DEV)
+        acc = torch.eye(4096, device=DEV)
+        for _ in range(30):
+            acc = acc @ x * 1e-3
+        m.weight.copy_(acc)
+        expect = acc.detach().cpu().clone()
+    eng.model_to_host(m)   # no torch.cuda.synchronize() first — by design
+    eng.synchronize()
+    assert torch.equal(m.weight.detach(), expect)
+
+
+def test_optimizer_state_offload_across_update():
+    """cfg.offload_optimizer path at 7B geometry: optimizer
+    state shuttled host↔device between updates must train identically to
+    keeping it resident."""
+    from nanorlhf_amd import ops
+    from nanorlhf_amd.models import CausalLM, get_config, pack_sequences
+    from nanorlhf_amd.utils.offload import OffloadEngine
+
+    def run(offload: bool):
+        torch.manual_seed(0)
+        cfg = get_config("qwen2.5-7b", num_layers=2, vocab_size=8192)
+        m = CausalLM(cfg).to(DEV).to(torch.bfloat16)
+        opt = ops.FusedAdamW(m.parameters(), lr=1e-3)
+        eng = OffloadEngine(torch.device(DEV), enabled=True)
+        torch.manual_seed(1)
+        for step in range(3):
+            ids, cu, mx, pos = pack_sequences([torch.randint(2, 8192, (48,))], device=DEV)
+            h = m(ids, CausalLM.train_ctx(cu, mx, pos))
+            lp, _ = ops.token_logprob_entropy(h, m.lm_head_weight,
+                                              torch.roll(ids, -1), 1.0)
+            (-lp.mean()).backward()
+            opt.step()
+            opt.zero_grad(set_to_none=True)
+            if offload:
+                eng.optimizer_state_to(opt, "cpu")
+                eng.synchronize()
+                assert all(v.device.type == "cpu"
+                           for st in opt.state.values()
+                           for v in st.values() if torch.is_tensor(v))
+                eng.optimizer_state_to(opt, DEV)
+                eng.join_compute()
+        return {n: p.detach().float().cpu() for n, p in m.named_parameters()}
+
+    base = run(False)
+    off = run(True)
+    for n in base:
+        assert torch.equal(base[n], off[n]), n
+
+
+def test_trainer_offload_optimizer_knob():
+    """offload_optimizer=True wired through a real trainer update (the
+    round-1 dead knob)."""
+    from nanorlhf_amd.algos import grpo
+    from nanorlhf_amd.algos.grpo import GRPOConfig
+    from nanorlhf_amd.data import hh_shaped_prompts
+    from nanorlhf_amd.models import CausalLM, get_config
+
+    torch.manual_seed(0)
+    cfg_m = get_config("qwen2.5-1.5b", num_layers=2, vocab_size=4096)
+    policy = CausalLM(cfg_m)
+    ref = CausalLM(cfg_m)
+    ref.load_state_dict(policy.state_dict())
+    cfg = GRPOConfig(model_preset="custom", dtype="bfloat16", use_lora=True,
+                     lora_r=8, per_device_train_batch_size=2,
+                     gradient_accumulation_steps=2, num_mini_batches=2,
+                     total_episodes=16, sample_n=2, response_length=16,
+                     temperature=1.0, stop_token_id=1,
+                     output_dir="/tmp/nanorlhf_gpu_offopt",
+                     score_token_budget=4096, offload_optimizer=True)
+    prompts = hh_shaped_prompts(16, 4096, min_len=8, max_len=24)
+    tr = grpo.make_trainer(cfg, policy, ref,
+                           lambda seqs: torch.tensor([float(len(s) % 3) for s in seqs]),
+                           prompts)
+    tr.train(num_updates=2)
+    assert tr.global_step == 2
+    m = tr._last_metrics
+    assert all(torch.isfinite(torch.tensor(float(v))) for k, v in m.items()
+               if isinstance(v, (int, float))), m
+
+
 def test_7b_geometry_small_depth():
     """Qwen2.5-7B geometry (28 q heads / 4 kv heads -> GQA G=7, untied
     lm_head) through rollout + fwd/bwd at reduced depth."""
