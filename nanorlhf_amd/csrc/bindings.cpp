@@ -30,6 +30,11 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor k_cache,
 // masked.hip
 std::vector<torch::Tensor> masked_moments(torch::Tensor v, torch::Tensor mask);
 torch::Tensor whiten_apply(torch::Tensor v, double mean, double invstd, double shift);
+// lora.hip
+torch::Tensor lora_gemm(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> u,
+                        c10::optional<torch::Tensor> b,
+                        c10::optional<torch::Tensor> bias);
 // attention.hip
 std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor cu_seqlens,
@@ -55,6 +60,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode);
   m.def("masked_moments", &masked_moments);
   m.def("whiten_apply", &whiten_apply);
+  m.def("lora_gemm", &lora_gemm, py::arg("x"), py::arg("w"),
+        py::arg("u") = py::none(), py::arg("b") = py::none(),
+        py::arg("bias") = py::none());
   m.def("fa_fwd_varlen", &fa_fwd_varlen);
   m.def("fa_bwd_varlen", &fa_bwd_varlen);
 }

@@ -40,6 +40,7 @@ class LoRALinear(nn.Module):
         self.scaling = alpha / r
         self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
         self._merged: torch.Tensor | None = None
+        self._weight_t: torch.Tensor | None = None  # Wᵀ cache for fused bwd
 
     @property
     def in_features(self):
@@ -60,8 +61,18 @@ class LoRALinear(nn.Module):
     def forward(self, x):
         if self._merged is not None and not torch.is_grad_enabled():
             return F.linear(x, self._merged, self.base.bias)
-        y = self.base(x)
         lx = self.dropout(x)
+        from ..ops.lora import fused_lora_linear, fused_path_ok
+        if (lx is x and fused_path_ok(self.base.weight, self.lora_A.shape[0])
+                and x.dtype == torch.bfloat16):
+            # single-launch fused HIP GEMM (csrc/lora.hip); Wᵀ cached once —
+            # the base weight is frozen while LoRA trains
+            if self._weight_t is None or self._weight_t.device != x.device:
+                self._weight_t = self.base.weight.detach().t().contiguous()
+            return fused_lora_linear(x, self.base.weight, self.base.bias,
+                                     self.lora_A, self.lora_B, self.scaling,
+                                     self._weight_t)
+        y = self.base(x)
         return y + F.linear(F.linear(lx, self.lora_A), self.lora_B) * self.scaling
 
     def merge_for_rollout(self):

@@ -18,7 +18,8 @@ CSRC = os.path.join(ROOT, "nanorlhf_amd", "csrc")
 sources = [
     os.path.join(CSRC, f)
     for f in ["bindings.cpp", "elementwise.hip", "logprob.hip", "adamw.hip",
-              "sampling.hip", "kvcache.hip", "attention.hip", "masked.hip"]
+              "sampling.hip", "kvcache.hip", "attention.hip", "masked.hip",
+              "lora.hip"]
 ]
 
 setup(

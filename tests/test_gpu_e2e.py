@@ -129,42 +129,53 @@ def test_offload_orders_after_pending_compute():
 
 
 def test_optimizer_state_offload_across_update():
-    """cfg.offload_optimizer path at 7B geometry: optimizer
-    state shuttled host↔device between updates must train identically to
-    keeping it resident."""
+    """cfg.offload_optimizer path at 7B geometry: the
+    host↔device shuttle of optimizer state must be a BITWISE round trip
+    (same run — cross-run bitwise equality is not guaranteed because some
+    backward kernels accumulate with atomics), and training must continue
+    finite afterwards."""
     from nanorlhf_amd import ops
     from nanorlhf_amd.models import CausalLM, get_config, pack_sequences
     from nanorlhf_amd.utils.offload import OffloadEngine
 
-    def run(offload: bool):
-        torch.manual_seed(0)
-        cfg = get_config("qwen2.5-7b", num_layers=2, vocab_size=8192)
-        m = CausalLM(cfg).to(DEV).to(torch.bfloat16)
-        opt = ops.FusedAdamW(m.parameters(), lr=1e-3)
-        eng = OffloadEngine(torch.device(DEV), enabled=True)
-        torch.manual_seed(1)
-        for step in range(3):
-            ids, cu, mx, pos = pack_sequences([torch.randint(2, 8192, (48,))], device=DEV)
-            h = m(ids, CausalLM.train_ctx(cu, mx, pos))
-            lp, _ = ops.token_logprob_entropy(h, m.lm_head_weight,
-                                              torch.roll(ids, -1), 1.0)
-            (-lp.mean()).backward()
-            opt.step()
-            opt.zero_grad(set_to_none=True)
-            if offload:
-                eng.optimizer_state_to(opt, "cpu")
-                eng.synchronize()
-                assert all(v.device.type == "cpu"
-                           for st in opt.state.values()
-                           for v in st.values() if torch.is_tensor(v))
-                eng.optimizer_state_to(opt, DEV)
-                eng.join_compute()
-        return {n: p.detach().float().cpu() for n, p in m.named_parameters()}
+    torch.manual_seed(0)
+    cfg = get_config("qwen2.5-7b", num_layers=2, vocab_size=8192)
+    m = CausalLM(cfg).to(DEV).to(torch.bfloat16)
+    opt = ops.FusedAdamW(m.parameters(), lr=1e-3)
+    eng = OffloadEngine(torch.device(DEV), enabled=True)
 
-    base = run(False)
-    off = run(True)
-    for n in base:
-        assert torch.equal(base[n], off[n]), n
+    def one_step():
+        ids, cu, mx, pos = pack_sequences([torch.randint(2, 8192, (48,))], device=DEV)
+        h = m(ids, CausalLM.train_ctx(cu, mx, pos))
+        lp, _ = ops.token_logprob_entropy(h, m.lm_head_weight,
+                                          torch.roll(ids, -1), 1.0)
+        (-lp.mean()).backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+
+    for rt in range(2):
+        one_step()
+        before = {(id(p), k): v.detach().clone()
+                  for p, st in opt.state.items()
+                  for k, v in st.items() if torch.is_tensor(v)}
+        eng.optimizer_state_to(opt, "cpu")
+        eng.synchronize()
+        assert all(v.device.type == "cpu"
+                   for st in opt.state.values()
+                   for v in st.values() if torch.is_tensor(v))
+        eng.optimizer_state_to(opt, DEV)
+        eng.join_compute()
+        torch.cuda.synchronize()
+        after = {(id(p), k): v
+                 for p, st in opt.state.items()
+                 for k, v in st.items() if torch.is_tensor(v)}
+        assert set(before) == set(after)
+        for key in before:
+            assert after[key].device.type == "cuda"
+            assert torch.equal(before[key], after[key].to(before[key].device)), key
+    one_step()
+    for p in m.parameters():
+        assert torch.isfinite(p.detach().float()).all()
 
 
 def test_trainer_offload_optimizer_knob():
