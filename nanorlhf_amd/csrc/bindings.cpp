@@ -35,6 +35,7 @@ torch::Tensor lora_gemm(torch::Tensor x, torch::Tensor w,
                         c10::optional<torch::Tensor> u,
                         c10::optional<torch::Tensor> b,
                         c10::optional<torch::Tensor> bias);
+void lora_add_(torch::Tensor y, torch::Tensor u, torch::Tensor b);
 // attention.hip
 std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor cu_seqlens,
@@ -63,6 +64,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lora_gemm", &lora_gemm, py::arg("x"), py::arg("w"),
         py::arg("u") = py::none(), py::arg("b") = py::none(),
         py::arg("bias") = py::none());
+  m.def("lora_add_", &lora_add_);
   m.def("fa_fwd_varlen", &fa_fwd_varlen);
   m.def("fa_bwd_varlen", &fa_bwd_varlen);
 }

@@ -195,3 +195,91 @@ torch::Tensor lora_gemm(torch::Tensor x, torch::Tensor w,
   HIP_CHECK_LAST();
   return y;
 }
+
+// ---------------------------------------------------------------------------
+// Rank-64 adapter epilogue: y += u·Bᵀ   (y:[M,N] bf16 in place, u:[M,64],
+// B:[N,64]).  Used when hipBLASLt wins the base GEMM (deep-K shapes): the
+// adapter contribution is ONE custom MFMA pass over y instead of two skinny
+// library GEMMs + an eager add (3 launches + an extra y round trip).
+// ---------------------------------------------------------------------------
+__launch_bounds__(256, 4)
+__global__ void lora_add_kernel(short* __restrict__ y,
+                                const short* __restrict__ u,
+                                const short* __restrict__ b,
+                                int M, int N, int grid_n) {
+  __shared__ short lds[2][128 * 64];
+  const int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    const int q = nwg >> 3, r = nwg & 7;
+    const int xcd = wg & 7, i = wg >> 3;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const int m0 = (wg / grid_n) * 128, n0 = (wg % grid_n) * 128;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    const int c = rnd * 256 + tid;
+    const int row = c >> 3, col = (c & 7) << 3;
+    GLDS16(u + (long)min(m0 + row, M - 1) * 64 + col, &lds[0][c << 3]);
+    GLDS16(b + (long)min(n0 + row, N - 1) * 64 + col, &lds[1][c << 3]);
+  }
+  __syncthreads();
+
+  const int fr = lane & 15, fk = (lane >> 4) << 3;
+  f32x4_l acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4_l{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    bf16x8_l af[4], bf[4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+      af[mf] = *reinterpret_cast<const bf16x8_l*>(
+          lds[0] + ((wm * 64 + mf * 16 + fr) << 6) + kk * 32 + fk);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+      bf[nf] = *reinterpret_cast<const bf16x8_l*>(
+          lds[1] + ((wn * 64 + nf * 16 + fr) << 6) + kk * 32 + fk);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = mfma16(af[mf], bf[nf], acc[mf][nf]);
+  }
+
+  const int crow0 = (lane >> 4) << 2;
+#pragma unroll
+  for (int nf = 0; nf < 4; ++nf) {
+    const int gn = n0 + wn * 64 + nf * 16 + fr;
+    if (gn >= N) continue;
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int gm = m0 + wm * 64 + mf * 16 + crow0 + reg;
+        if (gm < M) {
+          const long idx = (long)gm * N + gn;
+          y[idx] = f2bf(bf2f(y[idx]) + acc[mf][nf][reg]);
+        }
+      }
+  }
+}
+
+void lora_add_(torch::Tensor y, torch::Tensor u, torch::Tensor b) {
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kBFloat16 && y.dim() == 2);
+  TORCH_CHECK(y.is_contiguous() && u.is_contiguous() && b.is_contiguous());
+  const long M = y.size(0), N = y.size(1);
+  TORCH_CHECK(u.size(0) == M && u.size(1) == 64);
+  TORCH_CHECK(b.size(0) == N && b.size(1) == 64);
+  const int grid_m = (int)((M + 127) / 128), grid_n = (int)((N + 127) / 128);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(lora_add_kernel, dim3(grid_m * grid_n), dim3(256), 0, stream,
+                     (short*)y.data_ptr(), (const short*)u.data_ptr(),
+                     (const short*)b.data_ptr(), (int)M, (int)N, grid_n);
+  HIP_CHECK_LAST();
+}

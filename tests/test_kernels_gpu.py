@@ -397,3 +397,28 @@ def test_lora_linear_module_uses_fused_path():
     y.sum().backward()
     assert ll.lora_A.grad is not None and ll.lora_B.grad is not None
     assert torch.isfinite(x.grad.float()).all()
+
+
+def test_lora_add_inplace_vs_ref():
+    """lora_add_ (rank-64 adapter epilogue): y += u·Bᵀ in place."""
+    for (M, N) in [(256, 2048), (250, 1536), (130, 17920), (100, 100)]:
+        y0 = _mt(M, N, scale=0.5, seed=20)
+        u = _mt(M, 64, scale=0.5, seed=21)
+        b = _mt(N, 64, scale=0.05, seed=22)
+        y = y0.clone()
+        ops.ext().lora_add_(y, u, b)
+        want = (y0.float() + u.float() @ b.float().t())
+        assert rel_err(y, want) < 4e-2, (M, N, rel_err(y, want))
+
+
+def test_lora_autotune_cache_populates():
+    from nanorlhf_amd.ops import lora as L
+    x = _mt(256, 1536, seed=30)
+    w = _mt(2048, 1536, scale=0.05, seed=31)
+    u = _mt(256, 64, seed=32)
+    b = _mt(2048, 64, scale=0.05, seed=33)
+    L._TUNE.clear()
+    y = L._dispatch_gemm(x, w, u, b, None)
+    assert (2048, 1536) in L._TUNE
+    want = L.lora_gemm_ref(x, w, u, b)
+    assert rel_err(y, want) < 4e-2
