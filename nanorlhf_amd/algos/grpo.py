@@ -33,17 +33,23 @@ class GRPO(AlgoSpec):
         cfg: GRPOConfig = trainer.cfg
         n = ro.sample_n
         adv_seq = F.grpo_group_advantage(ro.scores, n)      # [B*n]
-        rows = list(range(ro.num_rows))
+        # one host transfer for the whole selection (the per-element
+        # float(adv_seq[i]) loop was one sync per row on GPU scores)
+        adv_cpu = adv_seq.detach().cpu()
         if cfg.sparse_filter:
             # sparse GRPO: drop zero-advantage samples (grpo_r1_trainer.py:565-568)
-            rows = [i for i in rows if float(adv_seq[i]) != 0.0]
+            rows = torch.nonzero(adv_cpu != 0.0, as_tuple=False).squeeze(1).tolist()
             if not rows:
                 rows = [0]
+        else:
+            rows = list(range(ro.num_rows))
         if getattr(cfg, "keep_one_of_n", True) and n > 1:
-            groups = sorted({i // n for i in rows})
+            by_group: dict[int, list[int]] = {}
+            for i in rows:
+                by_group.setdefault(i // n, []).append(i)
             keep = []
-            for gidx in groups:
-                members = [i for i in rows if i // n == gidx]
+            for gidx in sorted(by_group):
+                members = by_group[gidx]
                 pick = int(torch.randint(0, len(members), (1,), generator=trainer._keep_gen))
                 keep.append(members[pick])
             rows = keep

@@ -23,6 +23,11 @@ class PPOConfig(RLHFConfig):
     value_learning_rate: float = 3e-6
     whiten_rewards: bool = False
     whiten_advantages: bool = False  # reference default: PPO/ppo.py:166 advantage_whiten=False
+    # value-LoRA (reference PPO/ppo.py:141-159: value_use_lora=True, r=64)
+    value_use_lora: bool = True
+    value_lora_r: int = 64
+    value_lora_alpha: int = 16
+    value_lora_dropout: float = 0.0
 
 
 class PPO(AlgoSpec):
@@ -67,10 +72,21 @@ class PPO(AlgoSpec):
 
 def make_trainer(cfg, policy, ref_policy, reward_fn, train_prompts,
                  value_model=None, **kw) -> RLHFTrainer:
+    # value-LoRA (ppo.py:141-159): adapters on the critic backbone, score
+    # head + embeddings fully trained (value_modules_to_save)
+    if value_model is not None and getattr(cfg, "value_use_lora", False):
+        from ..models.lora import LoraConfig, apply_lora
+        apply_lora(value_model, LoraConfig(
+            r=cfg.value_lora_r, alpha=cfg.value_lora_alpha,
+            dropout=cfg.value_lora_dropout,
+            modules_to_save=("embed_tokens", "lm_head", "score")))
     t = RLHFTrainer(cfg, PPO(), policy, ref_policy, reward_fn, train_prompts,
                     value_model=value_model, **kw)
-    # separate value LR (ppo.py:118-119): second param group
-    if len(t.optimizer.param_groups) > 1 and hasattr(cfg, "value_learning_rate"):
-        t.optimizer.param_groups[1]["lr"] = cfg.value_learning_rate
-        t.optimizer.param_groups[1]["initial_lr"] = cfg.value_learning_rate
+    # separate value LR (ppo.py:118-119) on the value param groups
+    # (policy/value × decay/no-decay — ppo_trainer.py:341-402)
+    if hasattr(cfg, "value_learning_rate"):
+        for g in t.optimizer.param_groups:
+            if g.get("name", "").startswith("value"):
+                g["lr"] = cfg.value_learning_rate
+                g["initial_lr"] = cfg.value_learning_rate
     return t

@@ -142,15 +142,24 @@ class RLHFTrainer:
         self.sampler = SamplerEngine(self.policy, kv_pool_tokens=pool_tokens,
                                      kv_cache_dtype=getattr(cfg, "kv_cache_dtype", "bf16"))
 
-        # optimizer over trainable params (+ value model for PPO joint update)
+        # optimizer over trainable params (+ value model for PPO joint update).
+        # Reference parity (PPO/ppo_trainer.py:341-402): 4 groups —
+        # policy/value × decay/no-decay (biases + norm weights never decay).
         params = [p for p in self.policy.parameters() if p.requires_grad]
         self.value_params = []
         if self.value_model is not None:
             self.value_params = [p for p in self.value_model.parameters() if p.requires_grad]
+        groups = []
+        for model, tag in ((self.policy, "policy"), (self.value_model, "value")):
+            if model is None:
+                continue
+            decay, nodecay = _decay_split(model)
+            groups.append({"params": decay, "weight_decay": cfg.weight_decay,
+                           "name": tag + "_decay"})
+            groups.append({"params": nodecay, "weight_decay": 0.0,
+                           "name": tag + "_nodecay"})
         self.optimizer = ops.FusedAdamW(
-            [{"params": params},
-             *([{"params": self.value_params}] if self.value_params else [])],
-            lr=cfg.learning_rate, betas=(cfg.adam_beta1, cfg.adam_beta2),
+            groups, lr=cfg.learning_rate, betas=(cfg.adam_beta1, cfg.adam_beta2),
             eps=cfg.adam_eps, weight_decay=cfg.weight_decay)
         for g in self.optimizer.param_groups:
             g.setdefault("initial_lr", cfg.learning_rate)
@@ -648,6 +657,23 @@ class RLHFTrainer:
         return self.ckpt.save(self.global_step, self.episode, policy_state,
                               self.policy.cfg.to_dict(), self.optimizer, None,
                               value_state, getattr(self, "_last_metrics", None))
+
+
+def _decay_split(model: torch.nn.Module):
+    """HF get_decay_parameter_names semantics (PPO/ppo_trainer.py:351-352):
+    weight decay applies to every trainable param EXCEPT biases and
+    normalization weights."""
+    decay, nodecay = [], []
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        leaf = name.rsplit(".", 1)[-1]
+        is_norm = "norm" in name.rsplit(".", 2)[-2] if "." in name else False
+        if leaf == "bias" or is_norm:
+            nodecay.append(p)
+        else:
+            decay.append(p)
+    return decay, nodecay
 
 
 class _nullctx:

@@ -256,3 +256,38 @@ def test_callback_stop_and_oversize_prompt_error(tmp_path):
     eng = SamplerEngine(tr.policy, kv_pool_tokens=64, page_size=16)
     with pytest.raises(RuntimeError, match="KV pool"):
         eng.generate([[2] * 200], SamplingParams(n=1, max_tokens=50))
+
+
+def test_ppo_optimizer_group_parity(tmp_path):
+    """Reference PPO optimizer parity (PPO/ppo_trainer.py:341-402): 4 param
+    groups policy/value × decay/no-decay; biases + norm weights never decay;
+    value groups run at value_learning_rate; value-LoRA applied
+    (ppo.py:141-159)."""
+    cfg = _mk(PPOConfig, tmp_path, weight_decay=0.01, value_learning_rate=1e-5,
+              value_use_lora=True, value_lora_r=4, value_lora_alpha=8)
+    policy, ref = _models(7)
+    torch.manual_seed(7)
+    vm = ScalarHeadModel.from_preset("tiny", num_labels=1, bidirectional=False)
+    prompts = hh_shaped_prompts(8, 1024, min_len=4, max_len=10, seed=7)
+    tr = ppo.make_trainer(cfg, policy, ref, _varied_reward, prompts, value_model=vm)
+
+    names = [g.get("name") for g in tr.optimizer.param_groups]
+    assert names == ["policy_decay", "policy_nodecay", "value_decay", "value_nodecay"]
+    for g in tr.optimizer.param_groups:
+        if g["name"].endswith("_nodecay"):
+            assert g["weight_decay"] == 0.0
+        else:
+            assert g["weight_decay"] == 0.01
+        if g["name"].startswith("value"):
+            assert g["lr"] == 1e-5
+    # value-LoRA engaged: adapters exist, backbone frozen, score head trained
+    from nanorlhf_amd.models.lora import LoRALinear
+    assert any(isinstance(m, LoRALinear) for m in vm.modules())
+    assert vm.score.weight.requires_grad
+    assert not vm.model.layers[0].self_attn.qkv_proj.base.weight.requires_grad
+    # no-decay groups contain only biases/norm weights (by construction all
+    # 1-D here)
+    for g in tr.optimizer.param_groups:
+        if g["name"].endswith("_nodecay"):
+            assert all(p.dim() == 1 for p in g["params"])
+    tr.train(num_updates=1)
