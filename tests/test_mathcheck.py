@@ -1,0 +1,135 @@
+"""Table-driven adversarial tests of the math-answer verifier
+(breadth modeled on the reference's vendored stack —
+toolkit_for_MATH/latex_answer_check.py:52-236, parsing_lib.py:71-230 —
+re-implemented, not lifted)."""
+import pytest
+
+from nanorlhf_amd.rewards.mathcheck import (answers_equal, extract_boxed,
+                                            extract_math_answer,
+                                            parse_structured)
+
+# (pred, gold, expected)
+EQUAL_CASES = [
+    # --- plain numbers / formatting ---------------------------------------
+    ("42", "42", True),
+    ("42.0", "42", True),
+    ("+3", "3", True),
+    ("1,234,567", "1234567", True),
+    ("1,234.5", "1234.5", True),
+    ("0.5", "1/2", True),
+    ("007", "7", True),
+    ("42", "43", False),
+    ("3.14159", "3.1416", True),          # float rel 1e-3
+    ("3.14", "3.5", False),
+    ("-2", "2", False),
+    # --- fractions / radicals ---------------------------------------------
+    (r"\frac{3}{4}", "0.75", True),
+    (r"\dfrac{3}{4}", r"\frac{6}{8}", True),
+    (r"\tfrac12", "0.5", True),
+    (r"\frac{1}{2}", r"\frac{1}{3}", False),
+    (r"\frac{\sqrt{2}}{2}", r"\frac{1}{\sqrt{2}}", True),
+    (r"\sqrt{8}", r"2\sqrt{2}", True),
+    (r"\sqrt{2}", "1.41421", True),
+    (r"\sqrt{2}", "2", False),
+    (r"\sqrt[3]{27}", "3", True),
+    # --- mixed numbers ------------------------------------------------------
+    (r"2\frac{1}{2}", "5/2", True),
+    (r"2\frac{1}{2}", "1", False),         # NOT 2·(1/2)
+    ("3 1/4", "13/4", True),
+    # --- pi / symbolic ------------------------------------------------------
+    (r"\frac{\pi}{4}", "0.785398", True),
+    (r"2\pi", "6.28318", True),
+    (r"\pi^2", "9.8696", True),
+    ("x+1", "1+x", True),
+    ("(x+1)^2", "x^2+2x+1", True),
+    ("(x+1)^2", "x^2+1", False),
+    # --- percent / units / degrees -----------------------------------------
+    (r"50\%", "0.5", True),
+    (r"50\%", "50", True),
+    (r"50\%", "0.7", False),
+    ("5 cm", "5", True),
+    ("10 \\text{ meters}", "10", True),
+    ("90^\\circ", "90", True),
+    ("12 square units", "12", True),
+    ("\\$15", "15", True),
+    # --- equations ----------------------------------------------------------
+    ("x = 5", "5", True),
+    ("y = 2/4", "0.5", True),
+    ("x = 5", "6", False),
+    # --- scientific notation ------------------------------------------------
+    (r"1.2 \times 10^5", "120000", True),
+    (r"3 \times 10^{-2}", "0.03", True),
+    (r"1.2 \times 10^5", "12000", False),
+    # --- intervals ----------------------------------------------------------
+    ("[0, 5)", "[0,5)", True),
+    ("(0, 5)", "[0, 5]", False),           # open vs closed differs
+    (r"(-\infty, 3]", r"(-\infty,3]", True),
+    (r"(-\infty, 3]", r"(-\infty, 3)", False),
+    (r"(-\infty, 0) \cup (1, +\infty)", r"(-\infty,0)\cup(1,\infty)", True),
+    (r"(-\infty, 0) \cup (1, \infty)", r"(-\infty, 0) \cup (2, \infty)", False),
+    (r"[1, 2] \cup [3, 4]", "[1,2]", False),  # union arity differs
+    (r"[\frac{1}{2}, 1)", "[0.5, 1)", True),
+    # --- tuples (ordered) ---------------------------------------------------
+    ("(1, 2)", "(1,2)", True),
+    ("(1, 2)", "(2, 1)", False),
+    ("(1, 2, 3)", "(1, 2, 3)", True),
+    ("(1, 2, 3)", "(1, 2)", False),
+    ("(1/2, 0.25)", "(0.5, 1/4)", True),
+    # --- sets (unordered) ---------------------------------------------------
+    (r"\{1, 2\}", r"\{2, 1\}", True),
+    (r"\{1, 2\}", r"\{1, 3\}", False),
+    (r"\{1, 2, 2\}", r"\{1, 2\}", False),  # multiset cardinality respected
+    (r"\{\frac{1}{2}, 3\}", r"\{3, 0.5\}", True),
+    # --- matrices -----------------------------------------------------------
+    (r"\begin{pmatrix}1 & 2\\3 & 4\end{pmatrix}",
+     r"\begin{pmatrix}1&2\\3&4\end{pmatrix}", True),
+    (r"\begin{pmatrix}1 & 2\\3 & 4\end{pmatrix}",
+     r"\begin{bmatrix}1&2\\3&4\end{bmatrix}", True),   # env bracket cosmetic
+    (r"\begin{pmatrix}1 & 2\\3 & 4\end{pmatrix}",
+     r"\begin{pmatrix}1&2\\3&5\end{pmatrix}", False),
+    (r"\begin{pmatrix}1 & 2\end{pmatrix}",
+     r"\begin{pmatrix}1\\2\end{pmatrix}", False),       # row vs column
+    (r"\begin{pmatrix}\frac{1}{2} & 0\\0 & 1\end{pmatrix}",
+     r"\begin{pmatrix}0.5&0\\0&1\end{pmatrix}", True),
+    # --- text answers -------------------------------------------------------
+    (r"\text{yes}", "yes", True),
+    ("East", "east", False),               # case-sensitive by design
+    # --- adversarial / degenerate -------------------------------------------
+    ("", "5", False),
+    ("5", "", False),
+    (r"\frac{1}{0}", "oo", False),         # don't equate 1/0 blindly
+    ("x \\geq 5", "x \\leq 5", False),
+    ("0.999999", "1", True),               # within float rel-1e-3 tolerance
+]
+
+
+@pytest.mark.parametrize("pred,gold,want", EQUAL_CASES)
+def test_answers_equal_table(pred, gold, want):
+    assert answers_equal(pred, gold, sympy_timeout_s=3.0) is want, (pred, gold)
+
+
+def test_table_size_floor():
+    assert len(EQUAL_CASES) >= 50
+
+
+def test_extract_boxed_nested():
+    assert extract_boxed(r"so \boxed{\frac{1}{2}}") == r"\frac{1}{2}"
+    assert extract_boxed(r"\boxed{a{b}c}") == "a{b}c"
+    assert extract_boxed("no box") is None
+    # last box wins
+    assert extract_boxed(r"\boxed{1} then \boxed{2}") == "2"
+
+
+def test_extract_math_answer_fallback():
+    assert extract_math_answer("the answer is 42.") == "42"
+    assert extract_math_answer(r"thus \boxed{7}") == "7"
+    assert extract_math_answer("no numbers here") is None
+
+
+def test_parse_structured_kinds():
+    assert parse_structured("(1,2)").kind == "interval_union"
+    assert parse_structured("(1,2,3)").kind == "tuple"
+    assert parse_structured(r"\{1,2\}").kind == "set"
+    assert parse_structured(r"\begin{pmatrix}1\\2\end{pmatrix}").kind == "matrix"
+    assert parse_structured("42") is None
+    assert parse_structured("(x+1)*(x+2)") is None
