@@ -422,3 +422,117 @@ def test_lora_autotune_cache_populates():
     assert (2048, 1536) in L._TUNE
     want = L.lora_gemm_ref(x, w, u, b)
     assert rel_err(y, want) < 4e-2
+
+
+# --------------------------------------------------------------------------
+# bench-shape golden tests: the shapes the headline actually
+# runs — fa at seqlen 1500-2048 with the 1.5B GQA geometry, CE at the full
+# 151936 vocab, paged decode at batch 2048
+# --------------------------------------------------------------------------
+
+def test_fa_fwd_bench_shape_1500():
+    """fa fwd at the GRPO default rollout shape: 12 q-heads / 2 kv-heads,
+    d=128, seqlens to 2048 (fp32 torch reference, elementwise bound)."""
+    torch.manual_seed(0)
+    lens = [1500, 2048, 1024, 637]
+    T = sum(lens)
+    cu = torch.zeros(len(lens) + 1, dtype=torch.int32, device=DEV)
+    cu[1:] = torch.cumsum(torch.tensor(lens, dtype=torch.int32, device=DEV), 0)
+    q = _mt(T, 12, 128, seed=1)
+    k = _mt(T, 2, 128, seed=2)
+    v = _mt(T, 2, 128, seed=3)
+    o = ops.flash_attn_varlen(q, k, v, cu, max(lens), causal=True)
+    o_ref = _sdpa_ref(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), 128 ** -0.5, causal=True)
+    assert rel_err(o.cpu(), o_ref) < 4e-2, rel_err(o.cpu(), o_ref)
+
+
+def test_fa_bwd_bench_shape_1500():
+    torch.manual_seed(0)
+    lens = [1500, 731]
+    T = sum(lens)
+    cu = torch.zeros(len(lens) + 1, dtype=torch.int32, device=DEV)
+    cu[1:] = torch.cumsum(torch.tensor(lens, dtype=torch.int32, device=DEV), 0)
+    q = _mt(T, 12, 128, seed=1).requires_grad_(True)
+    k = _mt(T, 2, 128, seed=2).requires_grad_(True)
+    v = _mt(T, 2, 128, seed=3).requires_grad_(True)
+    o = ops.flash_attn_varlen(q, k, v, cu, max(lens), causal=True)
+    do = _mt(T, 12, 128, seed=4, scale=0.5)
+    o.backward(do)
+    from nanorlhf_amd.ops.attention import _sdpa_ref_autograd
+    qr = q.detach().cpu().requires_grad_(True)
+    kr = k.detach().cpu().requires_grad_(True)
+    vr = v.detach().cpu().requires_grad_(True)
+    orf = _sdpa_ref_autograd(qr, kr, vr, cu.cpu(), 128 ** -0.5, causal=True)
+    orf.backward(do.cpu())
+    # long-sequence bwd: mean error is the meaningful bound (max is bf16
+    # tail noise over 1500-term reductions)
+    for g, gr in ((q.grad, qr.grad), (k.grad, kr.grad), (v.grad, vr.grad)):
+        denom = gr.abs().max() + 1e-6
+        assert float((g.cpu() - gr).abs().mean() / denom) < 2e-3
+        assert float((g.cpu() - gr).abs().max() / denom) < 0.15
+
+
+def test_ce_full_vocab_error_bound():
+    """ce_rowstats at V=151936 (the real lm_head): bf16 logits vs an fp64
+    torch reference — logprob + entropy within fp32-accumulation bounds."""
+    torch.manual_seed(0)
+    R, V = 64, 151936
+    logits = _mt(R, V, scale=4.0, seed=5)
+    labels = torch.randint(0, V, (R,), device=DEV)
+    lp = torch.empty(R, device=DEV)
+    ent = torch.empty(R, device=DEV)
+    lse = torch.empty(R, device=DEV)
+    ops.ext().ce_rowstats(logits, labels, 1.0, lp, ent, lse)
+    lf = logits.double()
+    lse_ref = torch.logsumexp(lf, dim=-1)
+    lp_ref = lf[torch.arange(R, device=DEV), labels] - lse_ref
+    p = torch.softmax(lf, dim=-1)
+    ent_ref = lse_ref - (p * lf).sum(-1)
+    assert float((lp.double() - lp_ref).abs().max()) < 5e-3
+    assert float((ent.double() - ent_ref).abs().max()) < 5e-3
+
+
+def test_paged_decode_batch_2048():
+    """paged decode at rollout scale: 2048 concurrent sequences (the 512×4
+    GRPO batch), mixed lengths, 1.5B GQA geometry."""
+    torch.manual_seed(0)
+    B, HKV, HQ, D, ps = 2048, 2, 12, 128, 16
+    g = torch.Generator().manual_seed(0)
+    lens = torch.randint(1, 160, (B,), generator=g).tolist()
+    max_pages = (max(lens) + ps - 1) // ps
+    total_pages = sum((l + ps - 1) // ps for l in lens) + 1
+    kc = torch.zeros(total_pages, ps, HKV, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros(total_pages, HKV, D, ps, dtype=torch.bfloat16, device=DEV)
+    tables = torch.zeros(B, max_pages, dtype=torch.int32)
+    page = 0
+    all_slots, all_k, all_v = [], [], []
+    for b, L in enumerate(lens):
+        npages = (L + ps - 1) // ps
+        for i in range(npages):
+            tables[b, i] = page + i
+        for t in range(L):
+            all_slots.append((page + t // ps) * ps + t % ps)
+        page += npages
+        all_k.append(_mt(L, HKV, D, seed=100 + b))
+        all_v.append(_mt(L, HKV, D, seed=200 + b))
+    k = torch.cat(all_k)
+    v = torch.cat(all_v)
+    slots = torch.tensor(all_slots, dtype=torch.long, device=DEV)
+    ops.kv_append(k, v, slots, kc, vc)
+    q = _mt(B, HQ, D, seed=7)
+    sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    out = ops.paged_attn_decode(q, kc, vc, tables.to(DEV), sl, scale=D ** -0.5)
+    # spot-check 8 rows against a dense fp32 reference
+    starts = torch.zeros(B, dtype=torch.long)
+    acc = 0
+    for b, L in enumerate(lens):
+        starts[b] = acc
+        acc += L
+    for b in torch.randint(0, B, (8,), generator=g).tolist():
+        L = lens[b]
+        kk = k[starts[b]: starts[b] + L].float()
+        vv = v[starts[b]: starts[b] + L].float()
+        for h in range(HQ):
+            att = (q[b, h].float() @ kk[:, h // (HQ // HKV)].t()) * (D ** -0.5)
+            o_ref = torch.softmax(att, -1) @ vv[:, h // (HQ // HKV)]
+            assert rel_err(out[b, h], o_ref) < 4e-2, (b, h)
