@@ -52,6 +52,241 @@ DEVINL __bf16 f2bf16t(float f) {
   return u.b;
 }
 
+// pack two f32 into one dword of 2xbf16 (compiler emits v_cvt_pk_bf16_f32)
+DEVINL unsigned pack2bf(float lo, float hi) {
+  __hip_bfloat162 h2 = __float22bfloat162_rn(float2{lo, hi});
+  return *reinterpret_cast<unsigned*>(&h2);
+}
+
+// ======================================================= FORWARD (v2) ====
+// 8-wave structure (cdna_hip_programming.md Appendix B "fused attention
+// prefill" ladder): each of 8 waves owns QBLK=32 q rows; KV walks in
+// 64-token LDS tiles (double-buffered, XOR-swizzled, async-STAGE split).
+// QK^T is computed SWAPPED — S^T = mfma(K, Q^T) — so each lane holds the
+// whole 64-score P-row of ONE q row (split with its half-partner lane),
+// making the online softmax fully lane-local: 31 fmax + one
+// permlane32_swap per tile, no LDS round trip for P, defer-max rescale
+// (RESCALE_THRESHOLD=8; P bounded by e^8, bf16-safe; decision taken
+// BEFORE this tile's exponentiation — the textbook-safe order).
+// P→bf16 via v_cvt_pk packing + permlane32_swap builds the PV fragments
+// in registers, and PV is ALSO computed transposed — O^T = mfma(V^T, P^T)
+// — which lands each lane's O accumulator in its OWN q-row column, so the
+// rescale and the 1/s epilogue are lane-local too.
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(512, 1)
+void fa_fwd_kernel8(const short* __restrict__ q,
+                    const short* __restrict__ k,
+                    const short* __restrict__ v,
+                    const int* __restrict__ cu,
+                    short* __restrict__ o,
+                    float* __restrict__ lse,
+                    int Hq, int Hkv, float scale) {
+  constexpr int QBLK = 32;            // q rows per wave
+  constexpr int QTILE = 8 * QBLK;     // 256 per workgroup
+  constexpr int KVBLK = 64;
+  constexpr int NDS = D / 16;         // d-slots per QK^T chain
+  constexpr int NDT = D / 32;         // O d-tiles
+  constexpr float THR = 8.f;          // defer-max threshold
+  const int seq = blockIdx.y;
+  const int h = blockIdx.z;
+  const int kvh = h / (Hq / Hkv);
+  const int s0 = cu[seq], s1 = cu[seq + 1];
+  const int len = s1 - s0;
+  const int q0 = blockIdx.x * QTILE;
+  if (q0 >= len) return;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int col = lane & 31;          // this lane's q row within the wave blk
+  const int hi = lane >> 5;           // half-wave: which 8-kv slot of frags
+  const int qw0 = q0 + wid * QBLK;
+  const int qi = qw0 + col;           // lane's q row (sequence-local)
+  const bool row_ok = qi < len;
+  const bool wave_active = qw0 < len;
+
+  __shared__ short Kt[2][KVBLK * D];   // [tok][d], XOR-swizzled rows
+  __shared__ short Vt[2][D * KVBLK];   // [d][tok], XOR-swizzled rows
+
+  // Q fragments (B-operand of the swapped QK^T): lane holds
+  // Q[qi][16*ds + 8*hi + j], j=0..7 — contiguous d, straight from global.
+  bf16x8 qf[NDS];
+  {
+    const long base = ((long)(s0 + min(qi, len - 1)) * Hq + h) * D;
+#pragma unroll
+    for (int ds = 0; ds < NDS; ds++) {
+      s16x8 raw = *reinterpret_cast<const s16x8*>(q + base + 16 * ds + 8 * hi);
+      qf[ds] = *reinterpret_cast<bf16x8*>(&raw);
+    }
+  }
+
+  float m = -INFINITY, s_ = 0.f;
+  f32x16_t acc_o[NDT];
+#pragma unroll
+  for (int t = 0; t < NDT; t++) acc_o[t] = f32x16_t{};
+
+  const int kv_end = CAUSAL ? min(len, q0 + QTILE) : len;       // WG bound
+  const int kv_end_w = CAUSAL ? min(len, qw0 + QBLK) : len;     // wave bound
+
+  // async-STAGE split staging (issue-early / write-late): 16 B per thread
+  // per slice, 512 threads.
+  constexpr int CHUNKS = KVBLK * D / 8;          // 16-B chunks per K (or V) tile
+  constexpr int NSLICE = (CHUNKS + 511) / 512;
+  s16x8 pk_[NSLICE], pv_[NSLICE];
+  auto issue_tile_loads = [&](int kv0) {
+#pragma unroll
+    for (int sl = 0; sl < NSLICE; sl++) {
+      const int idx = threadIdx.x + sl * 512;
+      if (CHUNKS < 512 && idx >= CHUNKS) continue;
+      const int tok = idx / (D / 8);
+      const int d0 = (idx % (D / 8)) * 8;
+      const int kvi = min(kv0 + tok, len - 1);
+      const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
+      pk_[sl] = *reinterpret_cast<const s16x8*>(k + b);
+      pv_[sl] = *reinterpret_cast<const s16x8*>(v + b);
+    }
+  };
+  auto write_tile_lds = [&](int buf) {
+#pragma unroll
+    for (int sl = 0; sl < NSLICE; sl++) {
+      const int idx = threadIdx.x + sl * 512;
+      if (CHUNKS < 512 && idx >= CHUNKS) continue;
+      const int tok = idx / (D / 8);
+      const int d0 = (idx % (D / 8)) * 8;
+      *reinterpret_cast<s16x8*>(&Kt[buf][swz_idx<2 * D>(tok, d0 * 2)]) = pk_[sl];
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        Vt[buf][swz_idx<2 * KVBLK>(d0 + j, tok * 2)] = pv_[sl][j];
+    }
+  };
+
+  if (kv_end > 0) issue_tile_loads(0);
+  int buf = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK, buf ^= 1) {
+    __syncthreads();          // prior readers of `buf` are done
+    write_tile_lds(buf);
+    __syncthreads();          // tile visible to all waves
+    if (kv0 + KVBLK < kv_end) issue_tile_loads(kv0 + KVBLK);
+    if (!wave_active || kv0 >= kv_end_w) continue;
+
+    // ---- S^T = K · Q^T : two 32-kv subtiles, C col = own q row ----------
+    f32x16_t st[2];
+#pragma unroll
+    for (int sub = 0; sub < 2; sub++) {
+      st[sub] = f32x16_t{};
+#pragma unroll
+      for (int ds = 0; ds < NDS; ds++) {
+        s16x8 rk = *reinterpret_cast<const s16x8*>(
+            &Kt[buf][swz_idx<2 * D>(sub * 32 + col, (16 * ds + 8 * hi) * 2)]);
+        st[sub] = mfma32x32x16(*reinterpret_cast<bf16x8*>(&rk), qf[ds], st[sub]);
+      }
+    }
+    // ---- mask + scale: p[sub*16+r] is score for kv = kv0 + 32*sub +
+    //      crow(r,hi), crow = (r&3) + 8*(r>>2) + 4*hi, all for q row qi --
+    float p[32];
+    const bool full = row_ok && (kv0 + KVBLK <= len)
+                      && (!CAUSAL || kv0 + KVBLK - 1 <= qw0);
+    if (full) {
+#pragma unroll
+      for (int sub = 0; sub < 2; sub++)
+#pragma unroll
+        for (int r = 0; r < 16; r++) p[sub * 16 + r] = st[sub][r] * scale;
+    } else {
+#pragma unroll
+      for (int sub = 0; sub < 2; sub++)
+#pragma unroll
+        for (int r = 0; r < 16; r++) {
+          const int ki = kv0 + 32 * sub + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const bool ok = row_ok && (ki < len) && (!CAUSAL || ki <= qi);
+          p[sub * 16 + r] = ok ? st[sub][r] * scale : -INFINITY;
+        }
+    }
+    // ---- lane-local row max (+ half-partner combine) --------------------
+    float pmax = p[0];
+#pragma unroll
+    for (int i = 1; i < 32; i++) pmax = fmaxf(pmax, p[i]);
+    {
+      auto r2 = __builtin_amdgcn_permlane32_swap(__float_as_uint(pmax),
+                                                 __float_as_uint(pmax), false, false);
+      const float partner = __uint_as_float(hi ? r2[0] : r2[1]);
+      pmax = fmaxf(pmax, partner);
+    }
+    // ---- defer-max: rescale only when some row grew past THR ------------
+    float alpha = 1.f;
+    const bool need = row_ok && !(pmax - m <= THR);   // true on first tile (m=-inf)
+    if (__any(need)) {
+      const float mn = fmaxf(m, pmax);
+      alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+      m = mn;
+#pragma unroll
+      for (int t = 0; t < NDT; t++)
+#pragma unroll
+        for (int r = 0; r < 16; r++) acc_o[t][r] *= alpha;
+    }
+    // ---- P = exp(score - m), row sum ------------------------------------
+    float rs = 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; i++) {
+      p[i] = (p[i] > -INFINITY) ? __expf(p[i] - m) : 0.f;
+      rs += p[i];
+    }
+    {
+      auto r2 = __builtin_amdgcn_permlane32_swap(__float_as_uint(rs),
+                                                 __float_as_uint(rs), false, false);
+      rs += __uint_as_float(hi ? r2[0] : r2[1]);
+    }
+    s_ = s_ * alpha + rs;
+    // ---- P → bf16 PV fragments in registers (cvt_pk + permlane32_swap) --
+    // pa[ks] = P^T B-operand for kv slot ks: lane holds
+    // P[qi][16*ks + 8*hi + j], j=0..7.
+    bf16x8 pa[4];
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      const int b0 = ks * 8;  // p-index base of this 8-kv group (own half)
+      unsigned w01 = pack2bf(p[b0 + 0], p[b0 + 1]);
+      unsigned w23 = pack2bf(p[b0 + 2], p[b0 + 3]);
+      unsigned w45 = pack2bf(p[b0 + 4], p[b0 + 5]);
+      unsigned w67 = pack2bf(p[b0 + 6], p[b0 + 7]);
+      auto rA = __builtin_amdgcn_permlane32_swap(w01, w45, false, false);
+      auto rB = __builtin_amdgcn_permlane32_swap(w23, w67, false, false);
+      union { unsigned u[4]; bf16x8 f; } fr;
+      fr.u[0] = (unsigned)rA[0];
+      fr.u[1] = (unsigned)rB[0];
+      fr.u[2] = (unsigned)rA[1];
+      fr.u[3] = (unsigned)rB[1];
+      pa[ks] = fr.f;
+    }
+    // ---- O^T += V^T · P^T : NDT d-tiles × 4 kv slots --------------------
+#pragma unroll
+    for (int t = 0; t < NDT; t++) {
+#pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        s16x8 rv = *reinterpret_cast<const s16x8*>(
+            &Vt[buf][swz_idx<2 * KVBLK>(t * 32 + col, (16 * ks + 8 * hi) * 2)]);
+        acc_o[t] = mfma32x32x16(*reinterpret_cast<bf16x8*>(&rv), pa[ks], acc_o[t]);
+      }
+    }
+  }
+
+  if (!wave_active || !row_ok) return;
+  // ---- epilogue: O = O^T column / s, packed 8-B stores ------------------
+  const float inv = s_ > 0.f ? 1.f / s_ : 0.f;
+  const long obase = ((long)(s0 + qi) * Hq + h) * D;
+#pragma unroll
+  for (int t = 0; t < NDT; t++) {
+#pragma unroll
+    for (int g = 0; g < 4; g++) {
+      // regs 4g..4g+3 are d = t*32 + 8*g + 4*hi + (0..3) — contiguous
+      const int d0 = t * 32 + 8 * g + 4 * hi;
+      short pack[4];
+#pragma unroll
+      for (int j = 0; j < 4; j++) pack[j] = f2bf(acc_o[t][4 * g + j] * inv);
+      *reinterpret_cast<unsigned long long*>(o + obase + d0) =
+          *reinterpret_cast<const unsigned long long*>(pack);
+    }
+  }
+  if (hi == 0)
+    lse[(long)(s0 + qi) * Hq + h] = m + __logf(fmaxf(s_, 1e-30f));
+}
+
 // =========================================================== FORWARD =====
 // q [T,Hq,D] bf16, k/v [T,Hkv,D] bf16, cu [B+1] i32 -> o [T,Hq,D], lse [T,Hq] f32
 template <int D, bool CAUSAL>
@@ -460,11 +695,11 @@ std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
   if (T == 0) return {o, lse};
-  const int qtiles = (int)((max_seqlen + 63) / 64);
-  dim3 grid(qtiles, B, Hq), block(256);
+  const int qtiles = (int)((max_seqlen + 255) / 256);
+  dim3 grid(qtiles, B, Hq), block(512);
   auto stream = at::hip::getCurrentHIPStream();
 #define FWD_LAUNCH(DD, CC)                                                     \
-  hipLaunchKernelGGL((fa_fwd_kernel<DD, CC>), grid, block, 0, stream,          \
+  hipLaunchKernelGGL((fa_fwd_kernel8<DD, CC>), grid, block, 0, stream,         \
                      (const short*)q.data_ptr(), (const short*)k.data_ptr(),   \
                      (const short*)v.data_ptr(), cu_seqlens.data_ptr<int>(),   \
                      (short*)o.data_ptr(), lse.data_ptr<float>(), Hq, Hkv,     \
@@ -474,6 +709,35 @@ std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
   else if (D == 32) { if (causal) FWD_LAUNCH(32, true); else FWD_LAUNCH(32, false); }
   else TORCH_CHECK(false, "unsupported head_dim ", D);
 #undef FWD_LAUNCH
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+// round-1 4-wave forward kept for within-probe A/B timing only
+std::vector<torch::Tensor> fa_fwd_varlen_v1(torch::Tensor q, torch::Tensor k,
+                                            torch::Tensor v, torch::Tensor cu_seqlens,
+                                            long max_seqlen, double scale, bool causal) {
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 && q.is_contiguous());
+  const long T = q.size(0);
+  const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
+  const int B = cu_seqlens.size(0) - 1;
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
+  if (T == 0) return {o, lse};
+  const int qtiles = (int)((max_seqlen + 63) / 64);
+  dim3 grid(qtiles, B, Hq), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+#define FWD1_LAUNCH(DD, CC)                                                    \
+  hipLaunchKernelGGL((fa_fwd_kernel<DD, CC>), grid, block, 0, stream,          \
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),   \
+                     (const short*)v.data_ptr(), cu_seqlens.data_ptr<int>(),   \
+                     (short*)o.data_ptr(), lse.data_ptr<float>(), Hq, Hkv,     \
+                     (float)scale)
+  if (D == 128) { if (causal) FWD1_LAUNCH(128, true); else FWD1_LAUNCH(128, false); }
+  else if (D == 64) { if (causal) FWD1_LAUNCH(64, true); else FWD1_LAUNCH(64, false); }
+  else if (D == 32) { if (causal) FWD1_LAUNCH(32, true); else FWD1_LAUNCH(32, false); }
+  else TORCH_CHECK(false, "unsupported head_dim ", D);
+#undef FWD1_LAUNCH
   HIP_CHECK_LAST();
   return {o, lse};
 }
