@@ -86,7 +86,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
   constexpr int KVBLK = 64;
   constexpr int NDS = D / 16;         // d-slots per QK^T chain
   constexpr int NDT = D / 32;         // O d-tiles
-  constexpr float THR = 8.f;          // defer-max threshold
+  constexpr float THR = 11.54f;       // defer-max threshold (8 nats, log2 units)
   const int seq = blockIdx.y;
   const int h = blockIdx.z;
   const int kvh = h / (Hq / Hkv);
@@ -180,7 +180,10 @@ void fa_fwd_kernel8(const short* __restrict__ q,
       }
     }
     // ---- mask + scale: p[sub*16+r] is score for kv = kv0 + 32*sub +
-    //      crow(r,hi), crow = (r&3) + 8*(r>>2) + 4*hi, all for q row qi --
+    //      crow(r,hi), crow = (r&3) + 8*(r>>2) + 4*hi, all for q row qi.
+    //      Scores carry scale*log2(e) so the softmax runs in base-2
+    //      (v_exp_f32 directly; saves one VALU mul per element) ---------
+    const float sl2 = scale * 1.4426950408889634f;
     float p[32];
     const bool full = row_ok && (kv0 + KVBLK <= len)
                       && (!CAUSAL || kv0 + KVBLK - 1 <= qw0);
@@ -188,7 +191,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
 #pragma unroll
       for (int sub = 0; sub < 2; sub++)
 #pragma unroll
-        for (int r = 0; r < 16; r++) p[sub * 16 + r] = st[sub][r] * scale;
+        for (int r = 0; r < 16; r++) p[sub * 16 + r] = st[sub][r] * sl2;
     } else {
 #pragma unroll
       for (int sub = 0; sub < 2; sub++)
@@ -196,7 +199,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
         for (int r = 0; r < 16; r++) {
           const int ki = kv0 + 32 * sub + (r & 3) + 8 * (r >> 2) + 4 * hi;
           const bool ok = row_ok && (ki < len) && (!CAUSAL || ki <= qi);
-          p[sub * 16 + r] = ok ? st[sub][r] * scale : -INFINITY;
+          p[sub * 16 + r] = ok ? st[sub][r] * sl2 : -INFINITY;
         }
     }
     // ---- lane-local row max (+ half-partner combine) --------------------
@@ -214,7 +217,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
     const bool need = row_ok && !(pmax - m <= THR);   // true on first tile (m=-inf)
     if (__any(need)) {
       const float mn = fmaxf(m, pmax);
-      alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+      alpha = (m == -INFINITY) ? 0.f : exp2f(m - mn);
       m = mn;
 #pragma unroll
       for (int t = 0; t < NDT; t++)
@@ -225,7 +228,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
     float rs = 0.f;
 #pragma unroll
     for (int i = 0; i < 32; i++) {
-      p[i] = (p[i] > -INFINITY) ? __expf(p[i] - m) : 0.f;
+      p[i] = (p[i] > -INFINITY) ? exp2f(p[i] - m) : 0.f;
       rs += p[i];
     }
     {
@@ -283,8 +286,9 @@ void fa_fwd_kernel8(const short* __restrict__ q,
           *reinterpret_cast<const unsigned long long*>(pack);
     }
   }
-  if (hi == 0)
-    lse[(long)(s0 + qi) * Hq + h] = m + __logf(fmaxf(s_, 1e-30f));
+  if (hi == 0)   // m is in log2 units: lse = ln2 * (m + log2 s)
+    lse[(long)(s0 + qi) * Hq + h] =
+        0.6931471805599453f * (m + log2f(fmaxf(s_, 1e-30f)));
 }
 
 // =========================================================== FORWARD =====
