@@ -1,0 +1,108 @@
+"""Reward-curve validation of the fast rollout modes.
+
+Trains the same GRPO task three ways with fixed seeds:
+  A. baseline      — bf16 KV cache, logprobs recomputed by the scoring pass
+                     (the reference's behavior: its vLLM flow discards
+                     logprobs, grpo_trainer.py:536-577)
+  B. fp8 KV        — OCP e4m3 paged KV cache
+  C. fp8 + sampler-logprobs — also uses the sampler's own per-token
+                     logprobs as the behavior policy (skips the policy
+                     half of the scoring pass)
+
+Task: learnable synthetic reward (fraction of response tokens in a target
+id band) on a 4-layer Qwen2.5-1.5B-geometry model — the reward rises
+within ~12 updates, enough signal to see a training-quality divergence.
+
+Writes profiles/rollout_modes_r2.json with the per-update reward/KL
+curves and a verdict line.  Run on a GPU box:
+    python tools/validate_rollout_modes.py
+"""
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import torch
+
+from nanorlhf_amd.algos import grpo
+from nanorlhf_amd.data import hh_shaped_prompts
+from nanorlhf_amd.models import CausalLM, get_config
+
+UPDATES = 12
+VOCAB = 8192
+TARGET_LO, TARGET_HI = 2000, 3000
+
+
+def reward(seqs):
+    out = []
+    for s in seqs:
+        resp = s[-48:]
+        out.append(sum(1.0 for t in resp if TARGET_LO <= t < TARGET_HI) / max(1, len(resp)))
+    return torch.tensor(out) * 4.0
+
+
+def run(tag, kv_dtype, rollout_lp):
+    torch.manual_seed(0)
+    cfg_m = get_config("qwen2.5-1.5b", num_layers=4, vocab_size=VOCAB)
+    policy = CausalLM(cfg_m)
+    ref = CausalLM(cfg_m)
+    ref.load_state_dict(policy.state_dict())
+    cfg = grpo.GRPOConfig(
+        model_preset="custom", dtype="bfloat16", use_lora=True, lora_r=64,
+        lora_alpha=64, learning_rate=2e-4, lr_scheduler_type="constant",
+        per_device_train_batch_size=8, gradient_accumulation_steps=2,
+        num_mini_batches=2, total_episodes=10_000, sample_n=4,
+        response_length=48, temperature=1.0, top_p=0.95, stop_token_id=1,
+        kl_coef=0.02, output_dir=f"/tmp/rollout_modes_{tag}",
+        score_token_budget=16384, missing_eos_penalty=None,
+        kv_cache_dtype=kv_dtype, use_rollout_logprobs=rollout_lp,
+        gradient_checkpointing=False, save_steps=0, seed=1234)
+    prompts = hh_shaped_prompts(64, VOCAB, min_len=8, max_len=24, seed=7)
+    tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
+    curve = []
+    t0 = time.time()
+    for _ in range(UPDATES):
+        tr.train(num_updates=1)
+        m = tr._last_metrics
+        curve.append({"reward": m["objective/rlhf_reward_old"],
+                      "kl": m["objective/kl_old"],
+                      "ratio_var": m["val/ratio_var_new"]})
+    dt = time.time() - t0
+    print(f"[{tag}] {UPDATES} updates in {dt:.1f}s; rewards:",
+          [round(c["reward"], 3) for c in curve], flush=True)
+    return {"curve": curve, "seconds": dt}
+
+
+def main():
+    res = {}
+    res["baseline"] = run("baseline", "bf16", False)
+    res["fp8_kv"] = run("fp8_kv", "fp8_e4m3", False)
+    res["fp8_kv_sampler_lp"] = run("fp8lp", "fp8_e4m3", True)
+
+    def final(tag, k=4):
+        c = res[tag]["curve"]
+        return sum(x["reward"] for x in c[-k:]) / k
+
+    fb, f8, f8lp = final("baseline"), final("fp8_kv"), final("fp8_kv_sampler_lp")
+    # acceptance: final-reward within 10% of the baseline's learned gain
+    base_gain = fb - res["baseline"]["curve"][0]["reward"]
+    tol = max(0.1 * abs(base_gain), 0.05)
+    verdict = {
+        "fp8_kv_ok": bool(abs(f8 - fb) <= tol),
+        "sampler_lp_ok": bool(abs(f8lp - fb) <= tol),
+        "final_rewards": {"baseline": fb, "fp8_kv": f8, "fp8_kv_sampler_lp": f8lp},
+        "tolerance": tol,
+    }
+    res["verdict"] = verdict
+    out = os.path.join(os.path.dirname(__file__), "..", "gpurun_out",
+                       "rollout_modes_r2.json")
+    os.makedirs(os.path.dirname(out), exist_ok=True)
+    with open(out, "w") as f:
+        json.dump(res, f, indent=1)
+    print("VERDICT:", json.dumps(verdict))
+
+
+if __name__ == "__main__":
+    main()
