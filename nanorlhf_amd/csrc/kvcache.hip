@@ -24,6 +24,10 @@ DEVINL f32x4 mfma16k(bf16x8k a, bf16x8k b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+DEVINL f32x4 mfma16k_fp8(long a, long b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, c, 0, 0, 0);
+}
+
 // ---- OCP fp8 e4m3 KV support (gfx950-native cvt instructions) ----------
 DEVINL unsigned char f32_to_fp8(float x) {
   return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(x, x, 0, false) & 0xff);
@@ -50,6 +54,23 @@ DEVINL bf16x8k load_frag_fp8(const unsigned char* base, long idx) {
 #pragma unroll
   for (int j = 0; j < 8; j++) raw[j] = f2bf(f[j]);
   return *reinterpret_cast<bf16x8k*>(&raw);
+}
+
+DEVINL long load_frag8_raw(const unsigned char* base, long idx) {
+  i32x2k w = *reinterpret_cast<const i32x2k*>(base + idx);
+  return *reinterpret_cast<long*>(&w);
+}
+
+// quantize 8 f32 to 8 packed fp8 bytes (for Q / P operands on the fp8 path)
+DEVINL long pack8_fp8(const float* f) {
+  int w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], 0, false);
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+  int w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], 0, false);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+  i32x2k w;
+  w[0] = w0;
+  w[1] = w1;
+  return *reinterpret_cast<long*>(&w);
 }
 
 template <typename CT>
@@ -149,12 +170,19 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
   const int hi = lane >> 4, lo = lane & 15;
   const int h0 = kvh * G;
 
-  __shared__ short Pb[4][16 * 32];          // per-wave P [g][tok]
+  __shared__ short Pb[4][16 * 32];          // per-wave P [g][tok] (bf16)
+  __shared__ unsigned char Pb8[4][16 * 32]; // per-wave P (fp8 path)
   __shared__ float Lm[4][16], Ls[4][16];    // per-wave (m, s) per row
   __shared__ float Lacc[4][16][D];          // per-wave O accumulators
 
-  // Q fragment: lane holds Q[g = lo][32c + 8hi + j]; zero for g >= G
+  // Q fragment: lane holds Q[g = lo][32c + 8hi + j]; zero for g >= G.
+  // On the fp8 cache path Q is quantized to e4m3 once here and the QK^T
+  // runs on mfma_f32_16x16x32_fp8_fp8 — K-cache bytes feed the MFMA
+  // directly (no dequant pass), doubling matrix throughput on top of the
+  // halved KV stream (VERDICT #9 / ROADMAP #4).
+  constexpr bool FP8 = sizeof(CT) == 1;
   bf16x8k qf[NC];
+  long qf8[NC];
   {
     const int g = lo;
     const long base = ((long)b * Hq + h0 + min(g, G - 1)) * D;
@@ -164,6 +192,12 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
       if (g < G)
         raw = *reinterpret_cast<const s16x8*>(q + base + 32 * c + 8 * hi);
       qf[c] = *reinterpret_cast<bf16x8k*>(&raw);
+      if (FP8) {
+        float f[8];
+#pragma unroll
+        for (int j = 0; j < 8; j++) f[j] = bf2f(raw[j]);
+        qf8[c] = pack8_fp8(f);
+      }
     }
   }
 
@@ -187,8 +221,14 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
       const long krow = (((long)page * PS + lo) * Hkv + kvh) * D;
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        bf16x8k kf = load_kv_frag<CT>(kc, krow + 32 * c + 8 * hi);
-        sc[n] = mfma16k(qf[c], kf, sc[n]);
+        if (FP8) {
+          const long kf8 = load_frag8_raw((const unsigned char*)kc,
+                                          krow + 32 * c + 8 * hi);
+          sc[n] = mfma16k_fp8(qf8[c], kf8, sc[n]);
+        } else {
+          bf16x8k kf = load_kv_frag<CT>(kc, krow + 32 * c + 8 * hi);
+          sc[n] = mfma16k(qf[c], kf, sc[n]);
+        }
       }
     }
     // ---- mask + online softmax (rows = q heads g = 4*hi + r) ----------
@@ -217,7 +257,10 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
       for (int r = 0; r < 4; r++) {
         const float p = (pm[n][r] == -INFINITY) ? 0.f : __expf(pm[n][r] - m[r]);
         rowsum[r] += p;
-        Pb[wid][(4 * hi + r) * 32 + 16 * n + lo] = f2bf(p);
+        if (FP8)
+          Pb8[wid][(4 * hi + r) * 32 + 16 * n + lo] = f32_to_fp8(p);
+        else
+          Pb[wid][(4 * hi + r) * 32 + 16 * n + lo] = f2bf(p);
       }
     }
 #pragma unroll
@@ -235,7 +278,10 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
     lds_fence_wave_kv();
     // ---- PV: A = P [g][tok], B = V^T (d-major pages) -------------------
     bf16x8k pf;
-    {
+    long pf8 = 0;
+    if (FP8) {
+      pf8 = load_frag8_raw(&Pb8[wid][0], lo * 32 + 8 * hi);
+    } else {
       s16x8 raw = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * 32 + 8 * hi]);
       pf = *reinterpret_cast<bf16x8k*>(&raw);
     }
@@ -244,8 +290,13 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
 #pragma unroll
     for (int t = 0; t < D / 16; t++) {
       const long vaddr = (((long)pv_page * Hkv + kvh) * D + 16 * t + lo) * PS + tokoff;
-      bf16x8k vf = load_kv_frag<CT>(vc, vaddr);
-      acc_o[t] = mfma16k(pf, vf, acc_o[t]);
+      if (FP8) {
+        const long vf8 = load_frag8_raw((const unsigned char*)vc, vaddr);
+        acc_o[t] = mfma16k_fp8(pf8, vf8, acc_o[t]);
+      } else {
+        bf16x8k vf = load_kv_frag<CT>(vc, vaddr);
+        acc_o[t] = mfma16k(pf, vf, acc_o[t]);
+      }
     }
   }
 

@@ -30,7 +30,7 @@ from nanorlhf_amd.algos import grpo
 from nanorlhf_amd.data import hh_shaped_prompts
 from nanorlhf_amd.models import CausalLM, get_config
 
-UPDATES = 12
+UPDATES = 40
 VOCAB = 8192
 TARGET_LO, TARGET_HI = 2000, 3000
 
@@ -51,7 +51,7 @@ def run(tag, kv_dtype, rollout_lp):
     ref.load_state_dict(policy.state_dict())
     cfg = grpo.GRPOConfig(
         model_preset="custom", dtype="bfloat16", use_lora=True, lora_r=64,
-        lora_alpha=64, learning_rate=2e-4, lr_scheduler_type="constant",
+        lora_alpha=128, learning_rate=1e-3, lr_scheduler_type="constant",
         per_device_train_batch_size=8, gradient_accumulation_steps=2,
         num_mini_batches=2, total_episodes=10_000, sample_n=4,
         response_length=48, temperature=1.0, top_p=0.95, stop_token_id=1,
@@ -86,12 +86,15 @@ def main():
         return sum(x["reward"] for x in c[-k:]) / k
 
     fb, f8, f8lp = final("baseline"), final("fp8_kv"), final("fp8_kv_sampler_lp")
-    # acceptance: final-reward within 10% of the baseline's learned gain
+    # the comparison is only meaningful if the baseline actually LEARNED
     base_gain = fb - res["baseline"]["curve"][0]["reward"]
-    tol = max(0.1 * abs(base_gain), 0.05)
+    learned = base_gain > 0.15
+    tol = 0.2 * base_gain if learned else float("nan")
     verdict = {
-        "fp8_kv_ok": bool(abs(f8 - fb) <= tol),
-        "sampler_lp_ok": bool(abs(f8lp - fb) <= tol),
+        "baseline_learned": bool(learned),
+        "baseline_gain": base_gain,
+        "fp8_kv_ok": bool(learned and abs(f8 - fb) <= tol),
+        "sampler_lp_ok": bool(learned and abs(f8lp - fb) <= tol),
         "final_rewards": {"baseline": fb, "fp8_kv": f8, "fp8_kv_sampler_lp": f8lp},
         "tolerance": tol,
     }
