@@ -44,7 +44,10 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Ten
         return ext().paged_attn_decode(q.contiguous(), k_cache, v_cache,
                                        block_tables.contiguous(), seq_lens.contiguous(),
                                        float(scale))
-    # CPU reference
+    # CPU reference.  On an fp8 cache this mirrors the native-fp8 kernel's
+    # arithmetic (csrc/kvcache.hip): Q and the softmax P are ALSO e4m3 so
+    # the whole QK^T/PV runs on mfma_..._fp8_fp8 — quantize them here too.
+    fp8 = k_cache.dtype == torch.float8_e4m3fn
     B, Hq, D = q.shape
     page = k_cache.shape[1]
     Hkv = k_cache.shape[2]
@@ -53,13 +56,17 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Ten
     kc = k_cache.view(-1, Hkv, D).float()
     # V d-major [P, Hkv, D, ps] -> flat [P*ps, Hkv, D]
     vc = v_cache.permute(0, 3, 1, 2).reshape(-1, Hkv, D).float()
+
+    def q8(x):
+        return x.to(torch.float8_e4m3fn).float() if fp8 else x
+
     for b in range(B):
         L = int(seq_lens[b])
         pages = block_tables[b, : (L + page - 1) // page].long()
         slots = (pages.unsqueeze(1) * page + torch.arange(page)).reshape(-1)[:L]
         kk = kc[slots].repeat_interleave(rep, dim=1)  # [L, Hq, D]
         vv = vc[slots].repeat_interleave(rep, dim=1)
-        att = torch.einsum("hd,lhd->hl", q[b].float(), kk) * scale
-        att = torch.softmax(att, dim=-1)
+        att = torch.einsum("hd,lhd->hl", q8(q[b].float()), kk) * scale
+        att = q8(torch.softmax(att, dim=-1))
         out[b] = torch.einsum("hl,lhd->hd", att, vv)
     return out.to(q.dtype)
