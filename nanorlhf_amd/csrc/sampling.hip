@@ -142,6 +142,19 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   const float bin_lo = (float)bstar * (URANGE / NBINS) - URANGE;
   const float bin_hi = bin_lo + (URANGE / NBINS);
 
+  // refinement is only worth a full extra pass when the boundary bin
+  // holds real mass: keeping the WHOLE bin inflates the nucleus by at
+  // most hist[bstar], so when that is under a 2% slack just take
+  // u_thresh = bin_lo (the coarse histogram already places it within
+  // 1/32 nat) and skip pass 3 entirely (~20% of the kernel's traffic)
+  __shared__ int skip_refine_sh;
+  if (threadIdx.x == 0) {
+    const float bin_mass = hist[bstar];
+    skip_refine_sh = (bin_mass <= 0.02f) ? 1 : 0;
+    if (skip_refine_sh) { u_thresh_sh = bin_lo; found_sh = -1; }
+  }
+  __syncthreads();
+  if (!skip_refine_sh) {
   // ---- pass 3: refinement histogram inside bin* (membership by the SAME
   // coarse binning, so boundary values like u == 0 stay consistent) ------
   for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist[i] = 0.f;
@@ -176,6 +189,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     found_sh = -1;
   }
   __syncthreads();
+  }  // !skip_refine
   const float u_thresh = u_thresh_sh;
 
   // ---- pass 4: per-thread kept mass + block scan + owner walk ----------

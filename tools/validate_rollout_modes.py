@@ -31,6 +31,7 @@ from nanorlhf_amd.data import hh_shaped_prompts
 from nanorlhf_amd.models import CausalLM, get_config
 
 UPDATES = 40
+SEED = int(os.environ.get("VAL_SEED", "1234"))
 VOCAB = 8192
 TARGET_LO, TARGET_HI = 2000, 3000
 
@@ -58,7 +59,7 @@ def run(tag, kv_dtype, rollout_lp):
         kl_coef=0.02, output_dir=f"/tmp/rollout_modes_{tag}",
         score_token_budget=16384, missing_eos_penalty=None,
         kv_cache_dtype=kv_dtype, use_rollout_logprobs=rollout_lp,
-        gradient_checkpointing=False, save_steps=0, seed=1234)
+        gradient_checkpointing=False, save_steps=0, seed=SEED)
     prompts = hh_shaped_prompts(64, VOCAB, min_len=8, max_len=24, seed=7)
     tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
     curve = []
