@@ -88,46 +88,53 @@ def make_eval_fn(eval_prompts, gold):
 MODEL_PATH = None        # e.g. "/models/Qwen2-1.5B"
 DATA_JSONL = None        # e.g. "data/metamathqa.jsonl"
 
+def run_real(model_path: str, data_jsonl: str, cfg=None, num_updates=None):
+    """Real-data r1 training: HF checkpoint + tokenizer + (question, answer)
+    jsonl records with the rule reward over decoded strings (reference
+    grpo_r1.py:92,126-128,237-273)."""
+    import json as _json
+
+    from nanorlhf_amd.data.tokenizer import load_tokenizer, prepare_math_prompts
+    from nanorlhf_amd.models.hf_import import load_pretrained
+    from nanorlhf_amd.rewards import StringReward
+    from nanorlhf_amd.rewards.mathcheck import answers_equal, extract_boxed
+
+    cfg = cfg or config
+    policy = load_pretrained(model_path)
+    ref_policy = load_pretrained(model_path)
+    tokenizer = load_tokenizer(model_path)
+    cfg.stop_token_id = tokenizer.eos_token_id
+    cfg.pad_token_id = tokenizer.pad_token_id
+    with open(data_jsonl) as f:
+        records = [_json.loads(line) for line in f]
+    train_prompts, answer_map = prepare_math_prompts(records, tokenizer,
+                                                     max_prompt_len=512)
+
+    def rule_reward(texts, responses_ids, tok):
+        """Reference r1 contract (grpo_r1.py:250-273): boxed-answer
+        extraction + structured equivalence, gold looked up by the
+        prompt prefix of the decoded string."""
+        out = []
+        for t in texts:
+            gold = next((a for q, a in answer_map.items()
+                         if t.startswith(q)), None)
+            pred = extract_boxed(t)
+            ok = (gold is not None and pred is not None
+                  and answers_equal(pred, gold, sympy_timeout_s=0.015))
+            out.append(1.0 if ok else 0.0)
+        return torch.tensor(out)
+
+    reward_fn = StringReward(rule_reward, tokenizer, mode="r1")
+    trainer = grpo.make_trainer(cfg, policy, ref_policy, reward_fn,
+                                train_prompts)
+    trainer.train(num_updates=num_updates)
+    trainer.save()
+    return trainer
+
+
 if __name__ == "__main__":
     if MODEL_PATH is not None and DATA_JSONL is not None:
-        import json as _json
-
-        from nanorlhf_amd.data.tokenizer import load_tokenizer, prepare_math_prompts
-        from nanorlhf_amd.models.hf_import import load_pretrained
-        from nanorlhf_amd.rewards import StringReward
-        from nanorlhf_amd.rewards.mathcheck import answers_equal, extract_boxed
-
-        policy = load_pretrained(MODEL_PATH)
-        ref_policy = load_pretrained(MODEL_PATH)
-        tokenizer = load_tokenizer(MODEL_PATH)
-        config.stop_token_id = tokenizer.eos_token_id
-        config.pad_token_id = tokenizer.pad_token_id
-        with open(DATA_JSONL) as f:
-            records = [_json.loads(line) for line in f]
-        train_prompts, answer_map = prepare_math_prompts(records, tokenizer,
-                                                         max_prompt_len=512)
-        eval_records = records[: 64]
-        eval_prompts, eval_map = prepare_math_prompts(eval_records, tokenizer)
-
-        def rule_reward(texts, responses_ids, tok):
-            """Reference r1 contract (grpo_r1.py:250-273): boxed-answer
-            extraction + structured equivalence, gold looked up by the
-            prompt prefix of the decoded string."""
-            out = []
-            for t in texts:
-                gold = next((a for q, a in answer_map.items()
-                             if t.startswith(q)), None)
-                pred = extract_boxed(t)
-                ok = (gold is not None and pred is not None
-                      and answers_equal(pred, gold, sympy_timeout_s=0.015))
-                out.append(1.0 if ok else 0.0)
-            return torch.tensor(out)
-
-        reward_fn = StringReward(rule_reward, tokenizer, mode="r1")
-        trainer = grpo.make_trainer(config, policy, ref_policy, reward_fn,
-                                    train_prompts)
-        trainer.train()
-        trainer.save()
+        run_real(MODEL_PATH, DATA_JSONL)
         sys.exit(0)
 
     mcfg = get_config(config.model_preset)

@@ -295,3 +295,51 @@ def test_trainer_with_string_reward(tmp_path, tiny_tok):
                            prompts, device=torch.device("cpu"))
     tr.train(num_updates=1)
     assert tr.global_step == 1
+
+
+def test_r1_real_data_seam_end_to_end(tmp_path, tiny_tok):
+    """The r1 real-data path (examples/grpo_r1.py::run_real): tiny HF
+    checkpoint + tokenizer files + (question, answer) jsonl through one
+    sparse-GRPO update with the boxed-answer rule reward."""
+    import importlib.util
+    import json as _json
+
+    from nanorlhf_amd.models.config import ModelConfig
+    from nanorlhf_amd.models.hf_import import save_hf_checkpoint
+    from nanorlhf_amd.models.qwen2 import CausalLM as _CLM
+
+    # tiny HF-layout checkpoint with the tokenizer files beside it
+    cfg = ModelConfig(vocab_size=max(2048, len(tiny_tok) + 8), hidden_size=64,
+                      num_layers=2, num_heads=4, num_kv_heads=2, head_dim=16,
+                      intermediate_size=128, rope_theta=1e4, max_position=512,
+                      dtype="float32", tie_word_embeddings=True)
+    torch.manual_seed(0)
+    save_hf_checkpoint(_CLM(cfg), str(tmp_path / "ckpt"))
+    tiny_tok.save_pretrained(str(tmp_path / "ckpt"))
+
+    data = tmp_path / "math.jsonl"
+    with open(data, "w") as f:
+        for q, a in [("What is 2+2?", "4"), ("What is 3*3?", "9"),
+                     ("What is 10-4?", "6"), ("What is 1+1?", "2")]:
+            f.write(_json.dumps({"question": q, "answer": a}) + "\n")
+
+    spec = importlib.util.spec_from_file_location(
+        "grpo_r1_example",
+        os.path.join(os.path.dirname(__file__), "..", "examples", "grpo_r1.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    from nanorlhf_amd.algos.grpo import GRPOConfig
+    run_cfg = GRPOConfig(
+        output_dir=str(tmp_path / "out"), model_preset="tiny", dtype="float32",
+        use_lora=True, lora_r=4, lora_alpha=8, per_device_train_batch_size=2,
+        gradient_accumulation_steps=1, num_mini_batches=1, total_episodes=4,
+        sample_n=2, response_length=8, temperature=1.0, sparse_filter=True,
+        score_token_budget=2048, train_token_budget=0, kv_pool_tokens=4096,
+        gradient_checkpointing=False, save_steps=0, kl_coef=0.0,
+        missing_eos_penalty=None)
+    tr = mod.run_real(str(tmp_path / "ckpt"), str(data), cfg=run_cfg,
+                      num_updates=1)
+    assert tr.global_step == 1
+    # checkpoint written by trainer.save()
+    assert any(p.name.startswith("checkpoint-")
+               for p in (tmp_path / "out").iterdir())
