@@ -615,6 +615,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
     }
 
     // ---- P, dS (fp32, C layout) + stage P^T / dS / dS^T ----------------
+    // base-2 domain: P = 2^(S·scale·log2e − lse·log2e) (v_exp_f32 direct)
+    const float sl2 = scale * 1.4426950408889634f;
 #pragma unroll
     for (int n = 0; n < 2; n++) {
 #pragma unroll
@@ -625,7 +627,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
         float pv = 0.f, dsv = 0.f;
         if (ok) {
           const long li = (long)(s0 + qi) * Hq + h;
-          pv = __expf(sc[n][r] * scale - lse[li]);
+          pv = exp2f(fmaf(sc[n][r], sl2, -lse[li] * 1.4426950408889634f));
           dsv = pv * (dp[n][r] - drow[li]) * scale;
         }
         PT[wid][(16 * n + lo) * 16 + 4 * hi + r] = f2bf(pv);
