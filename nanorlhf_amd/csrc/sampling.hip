@@ -110,8 +110,12 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   const float M = sm[0];          // max of temperature-scaled logits
   const float invS = 1.f / ss[0];
 
-  // ---- pass 2: coarse histogram of prob mass ---------------------------
-  for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist[i] = 0.f;
+  // ---- pass 2: coarse histogram of prob mass (wave-privatized: 4 waves
+  // hammering one 1024-bin histogram serialize on LDS atomics; a private
+  // histogram per wave merged once removes the cross-wave conflicts) ----
+  __shared__ float hist4[SBLOCK / 64][NBINS];
+  for (int w = 0; w < SBLOCK / 64; w++)
+    for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist4[w][i] = 0.f;
   __syncthreads();
   for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
     s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
@@ -120,9 +124,12 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       const float u = bf2f(v[j]) * invT - M;
       int b = (int)((u + URANGE) * (NBINS / URANGE));
       b = max(0, min(NBINS - 1, b));
-      atomicAdd(&hist[b], __expf(u) * invS);
+      atomicAdd(&hist4[wid][b], __expf(u) * invS);
     }
   }
+  __syncthreads();
+  for (int i = threadIdx.x; i < NBINS; i += SBLOCK)
+    hist[i] = hist4[0][i] + hist4[1][i] + hist4[2][i] + hist4[3][i];
   __syncthreads();
   __shared__ float u_thresh_sh, mass_above_sh;
   __shared__ int bin_star_sh;
@@ -220,9 +227,13 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     // this thread owns the crossing: re-walk its strided chunks (the
     // sampled category order is thread-strided — a fixed permutation,
     // which leaves the sampled distribution exactly the kept softmax)
+    // scan ALL chunks with no data-dependent loop exit: an early-out on
+    // `found` makes every iteration's branch wait on the previous load
+    // (~594 serial HBM latencies); unconditional scan lets the loads
+    // pipeline and costs only the owner thread's single stride
     float acc = excl;
     int found = -1;
-    for (int i = threadIdx.x; i < nvec && found < 0; i += SBLOCK) {
+    for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
       s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
 #pragma unroll
       for (int j = 0; j < 8; j++) {
@@ -233,11 +244,11 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
         }
       }
     }
-    for (int i = nvec * 8 + threadIdx.x; i < V && found < 0; i += SBLOCK) {
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
       const float u = bf2f(lr[i]) * invT - M;
       if (u >= u_thresh) {
         acc += __expf(u) * invS;
-        if (target < acc) found = i;
+        if (target < acc && found < 0) found = i;
       }
     }
     if (found >= 0) atomicCAS(&found_sh, -1, found);
