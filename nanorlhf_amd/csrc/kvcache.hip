@@ -19,6 +19,7 @@
 
 typedef __bf16 bf16x8k __attribute__((ext_vector_type(8)));
 typedef int i32x2k __attribute__((ext_vector_type(2)));
+typedef int i32x4k __attribute__((ext_vector_type(4)));
 
 DEVINL f32x4 mfma16k(bf16x8k a, bf16x8k b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
@@ -112,10 +113,16 @@ __global__ void kv_append_kernel(const short* __restrict__ k,
     for (int j = 0; j < 8; j++)
       vc[((page * Hkv + h) * (long)D + d0 + j) * ps + off] = (CT)vv[j];
   } else {
-    // fp8 e4m3 quantized cache
+    // fp8 e4m3 quantized cache.  K is stored FRAGMENT-MAJOR within each
+    // token's D block: byte group (c = d0/32, hi = (d0%32)/8) lands at
+    // hi*(D/4) + c*8, so the decode QK^T reads its per-c 8-byte MFMA
+    // fragments as one contiguous 16-B load per c-PAIR (the natural
+    // [d] order would leave half-width 8-B loads on the K stream).
+    const int c = d0 >> 5, hi8 = (d0 >> 3) & 3;
+    const int kd0 = hi8 * (D >> 2) + c * 8;
 #pragma unroll
     for (int j = 0; j < 8; j++) {
-      kc[slot * (long)row_elems + e0 + j] = (CT)f32_to_fp8(bf2f(kk[j]));
+      kc[slot * (long)row_elems + h * D + kd0 + j] = (CT)f32_to_fp8(bf2f(kk[j]));
       vc[((page * Hkv + h) * (long)D + d0 + j) * ps + off] =
           (CT)f32_to_fp8(bf2f(vv[j]));
     }
@@ -219,13 +226,26 @@ __global__ void paged_decode_kernel(const short* __restrict__ q,    // [B, Hq, D
       sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
       const int page = btb[min(2 * tile + n, maxP - 1)];
       const long krow = (((long)page * PS + lo) * Hkv + kvh) * D;
+      if (FP8) {
+        // fragment-major K layout: this lane's NC fragments are NC*8
+        // contiguous bytes at hi*(D/4) — read them 16 B at a time
+        const long kbase = krow + hi * (D >> 2);
+        if (NC >= 2) {
 #pragma unroll
-      for (int c = 0; c < NC; c++) {
-        if (FP8) {
-          const long kf8 = load_frag8_raw((const unsigned char*)kc,
-                                          krow + 32 * c + 8 * hi);
-          sc[n] = mfma16k_fp8(qf8[c], kf8, sc[n]);
+          for (int c2 = 0; c2 < NC / 2; c2++) {
+            i32x4k w = *reinterpret_cast<const i32x4k*>(
+                (const unsigned char*)kc + kbase + c2 * 16);  // one b128 load
+            i32x2k lo2{w[0], w[1]}, hi2{w[2], w[3]};
+            sc[n] = mfma16k_fp8(qf8[2 * c2], *reinterpret_cast<const long*>(&lo2), sc[n]);
+            sc[n] = mfma16k_fp8(qf8[2 * c2 + 1], *reinterpret_cast<const long*>(&hi2), sc[n]);
+          }
         } else {
+          const long kf8 = load_frag8_raw((const unsigned char*)kc, kbase);
+          sc[n] = mfma16k_fp8(qf8[0], kf8, sc[n]);
+        }
+      } else {
+#pragma unroll
+        for (int c = 0; c < NC; c++) {
           bf16x8k kf = load_kv_frag<CT>(kc, krow + 32 * c + 8 * hi);
           sc[n] = mfma16k(qf[c], kf, sc[n]);
         }

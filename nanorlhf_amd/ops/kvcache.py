@@ -27,11 +27,38 @@ def kv_append(k: torch.Tensor, v: torch.Tensor, slots: torch.Tensor,
         return
     ps = k_cache.shape[1]
     kc = k_cache.view(-1, *k_cache.shape[2:])
-    kc[slots] = k
+    if k_cache.dtype == torch.float8_e4m3fn:
+        # mirror the kernel's fragment-major fp8 K layout
+        kc[slots] = k[..., _fp8_k_perm(k.shape[-1])].to(k_cache.dtype)
+    else:
+        kc[slots] = k
     pages = torch.div(slots, ps, rounding_mode="floor")
     offs = slots % ps
     # v_cache[page, h, d, off] = v[t, h, d]
     v_cache[pages, :, :, offs] = v
+
+
+def _fp8_k_perm(D: int) -> torch.Tensor:
+    """Fragment-major permutation for the fp8 K cache: stored index
+    hi*(D/4) + c*8 + j holds logical element c*32 + hi*8 + j."""
+    fwd = torch.empty(D, dtype=torch.long)
+    for d in range(D):
+        c, rem = divmod(d, 32)
+        hi, j = divmod(rem, 8)
+        fwd[hi * (D // 4) + c * 8 + j] = c * 32 + hi * 8 + j
+    return fwd
+
+
+def depermute_fp8_k(kc_float: torch.Tensor) -> torch.Tensor:
+    """Undo the fragment-major byte permutation of the fp8 K cache
+    (csrc/kvcache.hip kv_append)."""
+    D = kc_float.shape[-1]
+    idx = torch.empty(D, dtype=torch.long)
+    for d in range(D):
+        c, rem = divmod(d, 32)
+        hi, j = divmod(rem, 8)
+        idx[c * 32 + hi * 8 + j] = hi * (D // 4) + c * 8 + j
+    return kc_float[..., idx.to(kc_float.device)]
 
 
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
@@ -54,6 +81,8 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Ten
     rep = Hq // Hkv
     out = torch.empty_like(q, dtype=torch.float32)
     kc = k_cache.view(-1, Hkv, D).float()
+    if fp8:
+        kc = depermute_fp8_k(kc)
     # V d-major [P, Hkv, D, ps] -> flat [P*ps, Hkv, D]
     vc = v_cache.permute(0, 3, 1, 2).reshape(-1, Hkv, D).float()
 

@@ -257,9 +257,11 @@ def test_fp8_kv_append_and_paged_decode():
     slots_t = torch.tensor(slots, dtype=torch.long, device=DEV)
     ops.kv_append(k, v, slots_t, kc, vc)
     # hardware cvt vs torch cast: same value within one fp8 ulp
-    kc_flat = kc.view(-1, Hkv, D)
+    # (fp8 K pages are stored fragment-major — depermute before comparing)
+    from nanorlhf_amd.ops.kvcache import depermute_fp8_k
+    kc_flat = depermute_fp8_k(kc.view(-1, Hkv, D).float())
     want = k.float()
-    got = kc_flat[slots_t].float()
+    got = kc_flat[slots_t.cpu()] if not kc_flat.is_cuda else kc_flat[slots_t]
     assert float((got - want).abs().max() / want.abs().max()) < 0.08
     q = _mt(B, Hq, D, seed=9)
     sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
