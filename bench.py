@@ -49,6 +49,9 @@ def main():
                     action="store_false",
                     help="recompute behavior logprobs in the scoring pass "
                          "(the reference's behavior)")
+    ap.add_argument("--rollout-weight-dtype", type=str, default="bf16",
+                    help="bf16 (default) | fp8_e4m3 (e4m3 merged rollout "
+                         "weights; secondary measurement until promoted)")
     args = ap.parse_args()
 
     from nanorlhf_amd.algos import grpo
@@ -82,6 +85,7 @@ def main():
         score_token_budget=131072,
         train_token_budget=49152,
         kv_cache_dtype=args.kv_dtype,
+        rollout_weight_dtype=args.rollout_weight_dtype,
         use_rollout_logprobs=args.rollout_logprobs,
         output_dir=os.environ.get("BENCH_OUT", "/tmp/nanorlhf_bench"),
         save_steps=0, log_samples=0, report_to="none",

@@ -139,8 +139,10 @@ class RLHFTrainer:
 
         # sampler over the live policy
         pool_tokens = cfg.kv_pool_tokens or self._auto_pool_tokens(train_prompts)
-        self.sampler = SamplerEngine(self.policy, kv_pool_tokens=pool_tokens,
-                                     kv_cache_dtype=getattr(cfg, "kv_cache_dtype", "bf16"))
+        self.sampler = SamplerEngine(
+            self.policy, kv_pool_tokens=pool_tokens,
+            kv_cache_dtype=getattr(cfg, "kv_cache_dtype", "bf16"),
+            rollout_weight_dtype=getattr(cfg, "rollout_weight_dtype", "bf16"))
 
         # optimizer over trainable params (+ value model for PPO joint update).
         # Reference parity (PPO/ppo_trainer.py:341-402): 4 groups —

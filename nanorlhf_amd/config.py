@@ -84,6 +84,9 @@ class RLHFConfig:
     kv_pool_tokens: int = 0                # 0 → auto from batch & lengths
     kv_cache_dtype: str = "bf16"           # "fp8_e4m3": OCP fp8 paged KV (halves
                                            # the decode KV stream; opt-in)
+    rollout_weight_dtype: str = "bf16"     # "fp8_e4m3": e4m3 MERGED rollout
+                                           # weights via hipBLASLt fp8 GEMMs
+                                           # (training weights untouched)
     offload_ref: bool | None = None        # None → auto by memory pressure
     offload_reward: bool | None = None
     offload_optimizer: bool = False

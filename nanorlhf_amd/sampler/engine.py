@@ -41,7 +41,8 @@ class SamplerEngine:
     def __init__(self, model: CausalLM, kv_pool_tokens: int, page_size: int = 16,
                  max_num_seqs: int = 4096, prefill_chunk_tokens: int = 131072,
                  compact_interval: int = 32, use_graphs: bool = True,
-                 kv_cache_dtype: str = "bf16"):
+                 kv_cache_dtype: str = "bf16",
+                 rollout_weight_dtype: str = "bf16"):
         self.model = model
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
@@ -60,6 +61,12 @@ class SamplerEngine:
         self._graph_version = None
         self._step_dev = (torch.zeros(1, dtype=torch.long, device=self.device)
                           if self.device.type == "cuda" else None)
+        # "fp8_e4m3": rollout-only e4m3 layer weights via torch._scaled_mm
+        # (inference GEMMs at ~2x the bf16 rate; lm_head stays bf16 so the
+        # sampled logits/logprobs come from the exact head).  Training
+        # weights are never touched; validated like the fp8 KV mode.
+        self.rollout_weight_dtype = (rollout_weight_dtype
+                                     if self.device.type == "cuda" else "bf16")
 
     # ----------------------------------------------------------------- utils
     def _slots_for_range(self, seq: SeqState, start: int, end: int) -> list[int]:
@@ -173,7 +180,9 @@ class SamplerEngine:
         was_training = self.model.training
         self.model.eval()
         if merge_lora:
-            merge_for_rollout(self.model)
+            merge_for_rollout(self.model,
+                              quant=(self.rollout_weight_dtype
+                                     if self.rollout_weight_dtype != "bf16" else None))
         try:
             waiting: list[SeqState] = []
             uid = 0

@@ -44,7 +44,7 @@ def reward(seqs):
     return torch.tensor(out) * 4.0
 
 
-def run(tag, kv_dtype, rollout_lp):
+def run(tag, kv_dtype, rollout_lp, w_dtype="bf16"):
     torch.manual_seed(0)
     cfg_m = get_config("qwen2.5-1.5b", num_layers=4, vocab_size=VOCAB)
     policy = CausalLM(cfg_m)
@@ -59,6 +59,7 @@ def run(tag, kv_dtype, rollout_lp):
         kl_coef=0.02, output_dir=f"/tmp/rollout_modes_{tag}",
         score_token_budget=16384, missing_eos_penalty=None,
         kv_cache_dtype=kv_dtype, use_rollout_logprobs=rollout_lp,
+        rollout_weight_dtype=w_dtype,
         gradient_checkpointing=False, save_steps=0, seed=SEED)
     prompts = hh_shaped_prompts(64, VOCAB, min_len=8, max_len=24, seed=7)
     tr = grpo.make_trainer(cfg, policy, ref, reward, prompts)
@@ -81,12 +82,14 @@ def main():
     res["baseline"] = run("baseline", "bf16", False)
     res["fp8_kv"] = run("fp8_kv", "fp8_e4m3", False)
     res["fp8_kv_sampler_lp"] = run("fp8lp", "fp8_e4m3", True)
+    res["fp8_full"] = run("fp8full", "fp8_e4m3", True, w_dtype="fp8_e4m3")
 
     def final(tag, k=4):
         c = res[tag]["curve"]
         return sum(x["reward"] for x in c[-k:]) / k
 
     fb, f8, f8lp = final("baseline"), final("fp8_kv"), final("fp8_kv_sampler_lp")
+    f8full = final("fp8_full")
     # the comparison is only meaningful if the baseline actually LEARNED
     base_gain = fb - res["baseline"]["curve"][0]["reward"]
     learned = base_gain > 0.15
@@ -96,7 +99,9 @@ def main():
         "baseline_gain": base_gain,
         "fp8_kv_ok": bool(learned and abs(f8 - fb) <= tol),
         "sampler_lp_ok": bool(learned and abs(f8lp - fb) <= tol),
-        "final_rewards": {"baseline": fb, "fp8_kv": f8, "fp8_kv_sampler_lp": f8lp},
+        "fp8_full_ok": bool(learned and abs(f8full - fb) <= tol),
+        "final_rewards": {"baseline": fb, "fp8_kv": f8,
+                          "fp8_kv_sampler_lp": f8lp, "fp8_full": f8full},
         "tolerance": tol,
     }
     res["verdict"] = verdict
