@@ -5,6 +5,8 @@
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor invrms);
+std::vector<torch::Tensor> rmsnorm_addres_fwd(torch::Tensor x, torch::Tensor res,
+                                              torch::Tensor w, double eps);
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor table, torch::Tensor positions, double sign);
 torch::Tensor swiglu_fwd(torch::Tensor gu);
 torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu);
@@ -52,6 +54,7 @@ std::vector<torch::Tensor> fa_bwd_varlen(torch::Tensor dout, torch::Tensor q,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_addres_fwd", &rmsnorm_addres_fwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

@@ -288,3 +288,105 @@ torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu) {
   HIP_CHECK_LAST();
   return dgu;
 }
+
+// ----------------------------------------------- fused residual + RMSNorm
+// h = x + res;  y = rmsnorm(h) * w.  For H <= BLOCK*8 each thread keeps its
+// h slice in registers across both phases (read x,res; write h,y — the
+// unfused pair costs an extra h read plus a launch).  invrms saved for the
+// backward, which reuses rmsnorm_bwd on h (d h = d x = d res).
+template <int BLOCK>
+__global__ void rmsnorm_addres_fwd_kernel(const short* __restrict__ x,
+                                          const short* __restrict__ res,
+                                          const short* __restrict__ w,
+                                          short* __restrict__ h,
+                                          short* __restrict__ y,
+                                          float* __restrict__ invrms,
+                                          int H, float eps) {
+  __shared__ float scratch[BLOCK / 64];
+  const long row = blockIdx.x;
+  const short* xr = x + row * (long)H;
+  const short* rr = res + row * (long)H;
+  short* hr = h + row * (long)H;
+  short* yr = y + row * (long)H;
+  const int nvec = H / 8;
+  float ss = 0.f;
+  if (nvec <= BLOCK) {
+    // register-resident path (H <= BLOCK*8)
+    s16x8 hv{};
+    const int i = threadIdx.x;
+    if (i < nvec) {
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
+      s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float f = bf2f(xv[j]) + bf2f(rv[j]);
+        hv[j] = f2bf(f);
+        const float fb = bf2f(hv[j]);
+        ss += fb * fb;
+      }
+    }
+    ss = block_sum<BLOCK>(ss, scratch);
+    const float inv = rsqrtf(ss / (float)H + eps);
+    if (threadIdx.x == 0) invrms[row] = inv;
+    if (i < nvec) {
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float t = bf2f(f2bf(bf2f(hv[j]) * inv));
+        o[j] = f2bf(t * bf2f(wv[j]));
+      }
+      *reinterpret_cast<s16x8*>(hr + i * 8) = hv;
+      *reinterpret_cast<s16x8*>(yr + i * 8) = o;
+    }
+    return;
+  }
+  // general path: write h in pass 1, re-read (L2-hot) in pass 2
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i * 8);
+    s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i * 8);
+    s16x8 hv;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      hv[j] = f2bf(bf2f(xv[j]) + bf2f(rv[j]));
+      const float f = bf2f(hv[j]);
+      ss += f * f;
+    }
+    *reinterpret_cast<s16x8*>(hr + i * 8) = hv;
+  }
+  ss = block_sum<BLOCK>(ss, scratch);
+  const float inv = rsqrtf(ss / (float)H + eps);
+  if (threadIdx.x == 0) invrms[row] = inv;
+  for (int i = threadIdx.x; i < nvec; i += BLOCK) {
+    s16x8 hv = *reinterpret_cast<const s16x8*>(hr + i * 8);
+    s16x8 wv = *reinterpret_cast<const s16x8*>(w + i * 8);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      const float t = bf2f(f2bf(bf2f(hv[j]) * inv));
+      o[j] = f2bf(t * bf2f(wv[j]));
+    }
+    *reinterpret_cast<s16x8*>(yr + i * 8) = o;
+  }
+}
+
+std::vector<torch::Tensor> rmsnorm_addres_fwd(torch::Tensor x, torch::Tensor res,
+                                              torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && res.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.sizes() == res.sizes());
+  const long N = x.numel() / x.size(-1);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0);
+  auto h = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  auto invrms = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  if (N == 0) return {y, h, invrms};
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rmsnorm_addres_fwd_kernel<256>, dim3(N), dim3(256), 0, stream,
+                     (const short*)x.data_ptr(), (const short*)res.data_ptr(),
+                     (const short*)w.data_ptr(), (short*)h.data_ptr(),
+                     (short*)y.data_ptr(), invrms.data_ptr<float>(), H, (float)eps);
+  HIP_CHECK_LAST();
+  return {y, h, invrms};
+}

@@ -112,10 +112,16 @@ class Block(nn.Module):
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, rope_table, ctx):
-        x = x + self.self_attn(self.input_layernorm(x), rope_table, ctx)
-        x = x + self.mlp(self.post_attention_layernorm(x))
-        return x
+    def forward(self, delta, res, rope_table, ctx):
+        """Carries the residual stream as (delta, res) with hidden =
+        delta + res: the pending add fuses into the NEXT norm's kernel
+        (ops.add_rms_norm), so no standalone residual-add kernels run."""
+        eps = self.input_layernorm.eps
+        a_in, h = ops.add_rms_norm(delta, res, self.input_layernorm.weight, eps)
+        attn_delta = self.self_attn(a_in, rope_table, ctx)
+        m_in, h2 = ops.add_rms_norm(attn_delta, h,
+                                    self.post_attention_layernorm.weight, eps)
+        return self.mlp(m_in), h2
 
 
 class Transformer(nn.Module):
@@ -135,14 +141,16 @@ class Transformer(nn.Module):
         if self.rope_table.dtype != torch.float32:
             # keep the rope table fp32 even after model.to(bf16)
             self.rope_table = self.rope_table.float()
-        x = self.embed_tokens(input_ids)
+        delta = self.embed_tokens(input_ids)
+        res = None
         for layer in self.layers:
             if checkpoint and torch.is_grad_enabled():
-                x = torch.utils.checkpoint.checkpoint(layer, x, self.rope_table, ctx,
-                                                      use_reentrant=False)
+                delta, res = torch.utils.checkpoint.checkpoint(
+                    layer, delta, res, self.rope_table, ctx, use_reentrant=False)
             else:
-                x = layer(x, self.rope_table, ctx)
-        return self.norm(x)
+                delta, res = layer(delta, res, self.rope_table, ctx)
+        y, _ = ops.add_rms_norm(delta, res, self.norm.weight, self.norm.eps)
+        return y
 
 
 class CausalLM(nn.Module):

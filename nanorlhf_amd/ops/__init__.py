@@ -46,7 +46,7 @@ def ext_available() -> bool:
     return _load() is not None
 
 
-from .rmsnorm import rms_norm  # noqa: E402,F401
+from .rmsnorm import add_rms_norm, rms_norm  # noqa: E402,F401
 from .rope import build_rope_cache, rope_apply  # noqa: E402,F401
 from .swiglu import swiglu  # noqa: E402,F401
 from .attention import flash_attn_varlen  # noqa: E402,F401

@@ -540,3 +540,30 @@ def test_paged_decode_batch_2048():
             att = (q[b, h].float() @ kk[:, h // (HQ // HKV)].t()) * (D ** -0.5)
             o_ref = torch.softmax(att, -1) @ vv[:, h // (HQ // HKV)]
             assert rel_err(out[b, h], o_ref) < 4e-2, (b, h)
+
+
+def test_add_rms_norm_fused_fwd_bwd():
+    """Fused residual+RMSNorm vs the unfused composition (fwd y/h + dx,
+    dres, dw), both H<=2048 (register path) and H=4096 (general path)."""
+    for H in (1536, 4096):
+        torch.manual_seed(0)
+        x = _mt(300, H, scale=0.5, seed=1).requires_grad_(True)
+        r = _mt(300, H, scale=0.5, seed=2).requires_grad_(True)
+        w = (1 + 0.1 * _mt(H, seed=3).float()).to(torch.bfloat16).requires_grad_(True)
+        y, h = ops.add_rms_norm(x, r, w, 1e-6)
+        # h is also consumed downstream (residual chain)
+        loss = (y.float() * 0.3).sum() + (h.float() * 0.1).sum()
+        loss.backward()
+
+        x2 = x.detach().clone().requires_grad_(True)
+        r2 = r.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        h2 = x2 + r2
+        y2 = ops.rms_norm(h2, w2, 1e-6)
+        ((y2.float() * 0.3).sum() + (h2.float() * 0.1).sum()).backward()
+
+        assert rel_err(y, y2) < 2e-2, H
+        assert torch.equal(h.detach(), (x2 + r2).detach())
+        assert rel_err(x.grad, x2.grad) < 3e-2, H
+        assert rel_err(r.grad, r2.grad) < 3e-2, H
+        assert rel_err(w.grad, w2.grad) < 3e-2, H
