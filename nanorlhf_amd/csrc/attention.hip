@@ -291,201 +291,6 @@ void fa_fwd_kernel8(const short* __restrict__ q,
         0.6931471805599453f * (m + log2f(fmaxf(s_, 1e-30f)));
 }
 
-// =========================================================== FORWARD =====
-// q [T,Hq,D] bf16, k/v [T,Hkv,D] bf16, cu [B+1] i32 -> o [T,Hq,D], lse [T,Hq] f32
-template <int D, bool CAUSAL>
-__global__ void fa_fwd_kernel(const short* __restrict__ q,
-                              const short* __restrict__ k,
-                              const short* __restrict__ v,
-                              const int* __restrict__ cu,
-                              short* __restrict__ o,
-                              float* __restrict__ lse,
-                              int Hq, int Hkv, float scale) {
-  constexpr int QTILE = 64;   // per workgroup; 16 per wave
-  constexpr int KTILE = 64;   // KV tokens staged per iteration (4 subtiles)
-  constexpr int NSUB = KTILE / 16;
-  constexpr int NC = D / 32;  // K-chunks per mfma row
-  const int seq = blockIdx.y;
-  const int h = blockIdx.z;
-  const int kvh = h / (Hq / Hkv);
-  const int s0 = cu[seq], s1 = cu[seq + 1];
-  const int len = s1 - s0;
-  const int q0 = blockIdx.x * QTILE;
-  if (q0 >= len) return;
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int hi = lane >> 4;           // 0..3
-  const int lo = lane & 15;
-  const int qr0 = q0 + wid * 16;      // this wave's first q row (tile-local)
-
-  __shared__ short Kt[KTILE * D];       // row-major [tok][d], XOR-swizzled
-  __shared__ short Vt[D * KTILE];       // transposed [d][tok], XOR-swizzled
-  __shared__ short Pb[4][16 * KTILE];   // per-wave P [q][kv]
-
-  // ---- Q fragments: lane holds Q[lo][32c + 8hi + j] ---------------------
-  bf16x8 qf[NC];
-  const bool wave_active = qr0 < len;
-  {
-    const int qrow = qr0 + lo;
-    const long base = ((long)(s0 + min(qrow, len - 1)) * Hq + h) * D;
-#pragma unroll
-    for (int c = 0; c < NC; c++) {
-      s16x8 raw = *reinterpret_cast<const s16x8*>(q + base + 32 * c + 8 * hi);
-      qf[c] = *reinterpret_cast<bf16x8*>(&raw);
-    }
-  }
-
-  float m[4], s_[4];
-  f32x4 acc_o[D / 16];
-#pragma unroll
-  for (int r = 0; r < 4; r++) { m[r] = -INFINITY; s_[r] = 0.f; }
-#pragma unroll
-  for (int t = 0; t < D / 16; t++) acc_o[t] = f32x4{0.f, 0.f, 0.f, 0.f};
-
-  const int kv_end = CAUSAL ? min(len, q0 + QTILE) : len;
-  // async-STAGE split (cdna_hip_programming.md G15): each thread owns
-  // NSLICE 16-B pieces of the KV tile; the NEXT tile's global loads are
-  // issued before computing on the CURRENT tile so HBM latency hides under
-  // the MFMA phase.  Loads are unconditional with a clamped token index
-  // (branching per element would force a vmcnt(0) drain per load — guide
-  // §5.4 trap 4(c)); out-of-range tokens are neutralized by the score mask
-  // (P = 0), so garbage rows never contribute.
-  constexpr int NSLICE = KTILE * D / 8 / 256;
-  s16x8 pk_[NSLICE], pv_[NSLICE];
-  auto issue_tile_loads = [&](int kv0) {
-#pragma unroll
-    for (int sl = 0; sl < NSLICE; sl++) {
-      const int idx = threadIdx.x + sl * 256;
-      const int tok = idx / (D / 8);
-      const int d0 = (idx % (D / 8)) * 8;
-      const int kvi = min(kv0 + tok, len - 1);
-      const long b = ((long)(s0 + kvi) * Hkv + kvh) * D + d0;
-      pk_[sl] = *reinterpret_cast<const s16x8*>(k + b);
-      pv_[sl] = *reinterpret_cast<const s16x8*>(v + b);
-    }
-  };
-  auto write_tile_lds = [&]() {
-#pragma unroll
-    for (int sl = 0; sl < NSLICE; sl++) {
-      const int idx = threadIdx.x + sl * 256;
-      const int tok = idx / (D / 8);
-      const int d0 = (idx % (D / 8)) * 8;
-      *reinterpret_cast<s16x8*>(&Kt[swz_idx<2 * D>(tok, d0 * 2)]) = pk_[sl];
-#pragma unroll
-      for (int j = 0; j < 8; j++)
-        Vt[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = pv_[sl][j];
-    }
-  };
-  if (kv_end > 0) issue_tile_loads(0);
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KTILE) {
-    __syncthreads();
-    write_tile_lds();
-    __syncthreads();
-    if (kv0 + KTILE < kv_end) issue_tile_loads(kv0 + KTILE);
-    if (!wave_active) continue;
-
-    // ---- S = Q K^T : NSUB 16-col subtiles ------------------------------
-    f32x4 sc[NSUB];
-#pragma unroll
-    for (int n = 0; n < NSUB; n++) {
-      sc[n] = f32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int c = 0; c < NC; c++) {
-        s16x8 raw = *reinterpret_cast<const s16x8*>(
-            &Kt[swz_idx<2 * D>(16 * n + lo, 64 * c + 16 * hi)]);
-        sc[n] = mfma16x16x32(qf[c], *reinterpret_cast<bf16x8*>(&raw), sc[n]);
-      }
-    }
-    // ---- mask + online softmax ----------------------------------------
-    // C layout: row (q) = 4*hi + reg, col (kv) = lo + 16*n
-    float pm[NSUB][4];
-#pragma unroll
-    for (int n = 0; n < NSUB; n++) {
-#pragma unroll
-      for (int r = 0; r < 4; r++) {
-        const int qi = qr0 + 4 * hi + r;
-        const int ki = kv0 + 16 * n + lo;
-        const bool ok = (qi < len) && (ki < len) && (!CAUSAL || ki <= qi);
-        pm[n][r] = ok ? sc[n][r] * scale : -INFINITY;
-      }
-    }
-    float rowmax[4];
-#pragma unroll
-    for (int r = 0; r < 4; r++) {
-      float x = pm[0][r];
-#pragma unroll
-      for (int n = 1; n < NSUB; n++) x = fmaxf(x, pm[n][r]);
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off));
-      rowmax[r] = x;
-    }
-    float fac[4];
-#pragma unroll
-    for (int r = 0; r < 4; r++) {
-      const float mn = fmaxf(m[r], rowmax[r]);
-      fac[r] = (m[r] == -INFINITY) ? 0.f : __expf(m[r] - mn);
-      m[r] = mn;
-    }
-    // P = exp(score - m), row sums, stage P to per-wave LDS
-    float rowsum[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-    for (int n = 0; n < NSUB; n++) {
-#pragma unroll
-      for (int r = 0; r < 4; r++) {
-        const float p = (pm[n][r] == -INFINITY) ? 0.f : __expf(pm[n][r] - m[r]);
-        rowsum[r] += p;
-        Pb[wid][(4 * hi + r) * KTILE + 16 * n + lo] = f2bf(p);
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < 4; r++) {
-      float x = rowsum[r];
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) x += __shfl_xor(x, off);
-      s_[r] = s_[r] * fac[r] + x;
-    }
-    // rescale O accumulators (row = 4*hi + reg)
-#pragma unroll
-    for (int t = 0; t < D / 16; t++) {
-#pragma unroll
-      for (int r = 0; r < 4; r++) acc_o[t][r] *= fac[r];
-    }
-    lds_fence_wave();
-    // ---- PV: A = P (LDS), B = V^T slices; K=64 -> 2 mfma per d-tile ----
-    bf16x8 pf[2];
-    {
-      s16x8 r0 = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 8 * hi]);
-      s16x8 r1 = *reinterpret_cast<const s16x8*>(&Pb[wid][lo * KTILE + 32 + 8 * hi]);
-      pf[0] = *reinterpret_cast<bf16x8*>(&r0);
-      pf[1] = *reinterpret_cast<bf16x8*>(&r1);
-    }
-#pragma unroll
-    for (int t = 0; t < D / 16; t++) {
-      s16x8 rv0 = *reinterpret_cast<const s16x8*>(
-          &Vt[swz_idx<2 * KTILE>(t * 16 + lo, 16 * hi)]);
-      acc_o[t] = mfma16x16x32(pf[0], *reinterpret_cast<bf16x8*>(&rv0), acc_o[t]);
-      s16x8 rv1 = *reinterpret_cast<const s16x8*>(
-          &Vt[swz_idx<2 * KTILE>(t * 16 + lo, 64 + 16 * hi)]);
-      acc_o[t] = mfma16x16x32(pf[1], *reinterpret_cast<bf16x8*>(&rv1), acc_o[t]);
-    }
-  }
-
-  if (!wave_active) return;
-  // ---- epilogue: O /= s, store o + lse --------------------------------
-#pragma unroll
-  for (int r = 0; r < 4; r++) {
-    const int qi = qr0 + 4 * hi + r;
-    if (qi >= len) continue;
-    const float inv = s_[r] > 0.f ? 1.f / s_[r] : 0.f;
-    const long base = ((long)(s0 + qi) * Hq + h) * D;
-#pragma unroll
-    for (int t = 0; t < D / 16; t++)
-      o[base + t * 16 + lo] = f2bf(acc_o[t][r] * inv);
-    if (lo == 0)
-      lse[(long)(s0 + qi) * Hq + h] = m[r] + __logf(fmaxf(s_[r], 1e-30f));
-  }
-}
-
 // ====================================================== BWD: D = rowsum ==
 __global__ void fa_bwd_preprocess_kernel(const short* __restrict__ dout,
                                          const short* __restrict__ o,
@@ -715,35 +520,6 @@ std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
   else if (D == 32) { if (causal) FWD_LAUNCH(32, true); else FWD_LAUNCH(32, false); }
   else TORCH_CHECK(false, "unsupported head_dim ", D);
 #undef FWD_LAUNCH
-  HIP_CHECK_LAST();
-  return {o, lse};
-}
-
-// round-1 4-wave forward kept for within-probe A/B timing only
-std::vector<torch::Tensor> fa_fwd_varlen_v1(torch::Tensor q, torch::Tensor k,
-                                            torch::Tensor v, torch::Tensor cu_seqlens,
-                                            long max_seqlen, double scale, bool causal) {
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 && q.is_contiguous());
-  const long T = q.size(0);
-  const int Hq = q.size(1), D = q.size(2), Hkv = k.size(1);
-  const int B = cu_seqlens.size(0) - 1;
-  auto o = torch::empty_like(q);
-  auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
-  if (T == 0) return {o, lse};
-  const int qtiles = (int)((max_seqlen + 63) / 64);
-  dim3 grid(qtiles, B, Hq), block(256);
-  auto stream = at::hip::getCurrentHIPStream();
-#define FWD1_LAUNCH(DD, CC)                                                    \
-  hipLaunchKernelGGL((fa_fwd_kernel<DD, CC>), grid, block, 0, stream,          \
-                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),   \
-                     (const short*)v.data_ptr(), cu_seqlens.data_ptr<int>(),   \
-                     (short*)o.data_ptr(), lse.data_ptr<float>(), Hq, Hkv,     \
-                     (float)scale)
-  if (D == 128) { if (causal) FWD1_LAUNCH(128, true); else FWD1_LAUNCH(128, false); }
-  else if (D == 64) { if (causal) FWD1_LAUNCH(64, true); else FWD1_LAUNCH(64, false); }
-  else if (D == 32) { if (causal) FWD1_LAUNCH(32, true); else FWD1_LAUNCH(32, false); }
-  else TORCH_CHECK(false, "unsupported head_dim ", D);
-#undef FWD1_LAUNCH
   HIP_CHECK_LAST();
   return {o, lse};
 }

@@ -39,9 +39,6 @@ torch::Tensor lora_gemm(torch::Tensor x, torch::Tensor w,
                         c10::optional<torch::Tensor> bias);
 void lora_add_(torch::Tensor y, torch::Tensor u, torch::Tensor b);
 // attention.hip
-std::vector<torch::Tensor> fa_fwd_varlen_v1(torch::Tensor q, torch::Tensor k,
-                                            torch::Tensor v, torch::Tensor cu_seqlens,
-                                            long max_seqlen, double scale, bool causal);
 std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor cu_seqlens,
                                          long max_seqlen, double scale, bool causal);
@@ -72,6 +69,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = py::none());
   m.def("lora_add_", &lora_add_);
   m.def("fa_fwd_varlen", &fa_fwd_varlen);
-  m.def("fa_fwd_varlen_v1", &fa_fwd_varlen_v1);
   m.def("fa_bwd_varlen", &fa_bwd_varlen);
 }
