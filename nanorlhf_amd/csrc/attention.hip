@@ -333,14 +333,23 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
   const int hi = lane >> 4, lo = lane & 15;
   const int hi5 = lane >> 5, lo5 = lane & 31;
 
-  __shared__ short Kt[KTILE * D];     // [tok][d]
-  __shared__ short KT[D * KTILE];     // [d][tok]
+  // Staging layouts are FRAGMENT-MAJOR: each MFMA B/A fragment's 8
+  // contiguous elements are one 16-B LDS slot and the 32 (or 16) lanes of
+  // a fragment read CONSECUTIVE slots — the round-1 [row][col] images put
+  // 16-B reads at 32-B row strides, a 4-way bank conflict on every read
+  // (PMC: SQ_LDS_BANK_CONFLICT/SQ_LDS_IDX_ACTIVE = 0.505).
+  //   KT  [kv>>3][d][kv&7]      — dQ's B operand (8 kv per d row)
+  //   QT/dOT [c=d>>5][q>>3][d&31][q&7] — dK/dV's B operands
+  //   PT/dST [q>>3][kv][q&7]    — dV/dK's A operands
+  //   dSb [kv>>3][q][kv&7]      — dQ's A operand
+  __shared__ short Kt[KTILE * D];     // [tok][d] (row-contig reads, swizzled)
+  __shared__ short KT[4 * D * 8];     // frag-major, dQ B operand
   __shared__ short Vt[KTILE * D];     // [tok][d]
-  __shared__ short QT[4][D * 16];     // per-wave Q^T [d][q]
-  __shared__ short dOT[4][D * 16];    // per-wave dO^T [d][q]
-  __shared__ short PT[4][KTILE * 16];   // per-wave P^T [kv][q]
-  __shared__ short dSb[4][16 * KTILE];  // per-wave dS [q][kv]
-  __shared__ short dST[4][KTILE * 16];  // per-wave dS^T [kv][q]
+  __shared__ short QT[4][D * 16];     // per-wave, frag-major
+  __shared__ short dOT[4][D * 16];    // per-wave, frag-major
+  __shared__ short PT[4][KTILE * 16];   // per-wave, frag-major
+  __shared__ short dSb[4][16 * KTILE];  // per-wave, frag-major
+  __shared__ short dST[4][KTILE * 16];  // per-wave, frag-major
 
   // cooperative stage of K (both layouts) and V
   for (int idx = threadIdx.x; idx < KTILE * D / 8; idx += 256) {
@@ -357,7 +366,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
     *reinterpret_cast<s16x8*>(&Vt[swz_idx<2 * D>(tok, d0 * 2)]) = vv;
 #pragma unroll
     for (int j = 0; j < 8; j++)
-      KT[swz_idx<2 * KTILE>(d0 + j, tok * 2)] = kk[j];
+      KT[(((tok >> 3) * D) + d0 + j) * 8 + (tok & 7)] = kk[j];
   }
   __syncthreads();
 
@@ -396,8 +405,10 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
         dofr[c] = *reinterpret_cast<bf16x8*>(&rd);
 #pragma unroll
         for (int j = 0; j < 8; j++) {
-          QT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rq[j];
-          dOT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rd[j];
+          const int dd = 32 * c + 8 * hi + j;
+          const int off = ((c * 2 + (lo >> 3)) * 32 + (dd & 31)) * 8 + (lo & 7);
+          QT[wid][off] = rq[j];
+          dOT[wid][off] = rd[j];
         }
       }
     }
@@ -435,25 +446,30 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
           pv = exp2f(fmaf(sc[n][r], sl2, -lse[li] * 1.4426950408889634f));
           dsv = pv * (dp[n][r] - drow[li]) * scale;
         }
-        PT[wid][(16 * n + lo) * 16 + 4 * hi + r] = f2bf(pv);
-        dSb[wid][(4 * hi + r) * KTILE + 16 * n + lo] = f2bf(dsv);
-        dST[wid][(16 * n + lo) * 16 + 4 * hi + r] = f2bf(dsv);
+        {
+          const int qq = 4 * hi + r, kv = 16 * n + lo;
+          const int pt_off = ((qq >> 3) * KTILE + kv) * 8 + (qq & 7);
+          PT[wid][pt_off] = f2bf(pv);
+          dST[wid][pt_off] = f2bf(dsv);
+          dSb[wid][((kv >> 3) * 16 + qq) * 8 + (kv & 7)] = f2bf(dsv);
+        }
       }
     }
     lds_fence_wave();
 
     // ---- dV += P^T dO ; dK += dS^T Q (32x32x16, K = 16 q rows) ---------
     {
-      s16x8 rp = *reinterpret_cast<const s16x8*>(&PT[wid][lo5 * 16 + 8 * hi5]);
-      s16x8 rs = *reinterpret_cast<const s16x8*>(&dST[wid][lo5 * 16 + 8 * hi5]);
+      s16x8 rp = *reinterpret_cast<const s16x8*>(
+          &PT[wid][(hi5 * KTILE + lo5) * 8]);
+      s16x8 rs = *reinterpret_cast<const s16x8*>(
+          &dST[wid][(hi5 * KTILE + lo5) * 8]);
       bf16x8 pA = *reinterpret_cast<bf16x8*>(&rp);
       bf16x8 sA = *reinterpret_cast<bf16x8*>(&rs);
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        s16x8 rdo = *reinterpret_cast<const s16x8*>(
-            &dOT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
-        s16x8 rqt = *reinterpret_cast<const s16x8*>(
-            &QT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
+        const int boff = ((c * 2 + hi5) * 32 + lo5) * 8;
+        s16x8 rdo = *reinterpret_cast<const s16x8*>(&dOT[wid][boff]);
+        s16x8 rqt = *reinterpret_cast<const s16x8*>(&QT[wid][boff]);
         acc_dv[c] = mfma32x32x16(pA, *reinterpret_cast<bf16x8*>(&rdo), acc_dv[c]);
         acc_dk[c] = mfma32x32x16(sA, *reinterpret_cast<bf16x8*>(&rqt), acc_dk[c]);
       }
@@ -461,12 +477,13 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
 
     // ---- dQ = dS K (16x16x32 over d tiles) -----------------------------
     {
-      s16x8 rds = *reinterpret_cast<const s16x8*>(&dSb[wid][lo * KTILE + 8 * hi]);
+      s16x8 rds = *reinterpret_cast<const s16x8*>(
+          &dSb[wid][(hi * 16 + lo) * 8]);
       bf16x8 dsA = *reinterpret_cast<bf16x8*>(&rds);
 #pragma unroll
       for (int t = 0; t < D / 16; t++) {
         s16x8 rkt = *reinterpret_cast<const s16x8*>(
-            &KT[swz_idx<2 * KTILE>(t * 16 + lo, 16 * hi)]);
+            &KT[(hi * D + t * 16 + lo) * 8]);
         f32x4 dq = mfma16x16x32(dsA, *reinterpret_cast<bf16x8*>(&rkt), f32x4{0.f, 0.f, 0.f, 0.f});
 #pragma unroll
         for (int r = 0; r < 4; r++) {
