@@ -339,7 +339,9 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
   // 16-B reads at 32-B row strides, a 4-way bank conflict on every read
   // (PMC: SQ_LDS_BANK_CONFLICT/SQ_LDS_IDX_ACTIVE = 0.505).
   //   KT  [kv>>3][d][kv&7]      — dQ's B operand (8 kv per d row)
-  //   QT/dOT [c=d>>5][q>>3][d&31][q&7] — dK/dV's B operands
+  //   QT/dOT keep the [d][q] row image (scalar-transpose writes outnumber
+  //   the b128 reads 8:1 there — frag-major measured SLOWER from 8-way
+  //   write conflicts; A/B 2.32→3.01 ms)
   //   PT/dST [q>>3][kv][q&7]    — dV/dK's A operands
   //   dSb [kv>>3][q][kv&7]      — dQ's A operand
   __shared__ short Kt[KTILE * D];     // [tok][d] (row-contig reads, swizzled)
@@ -405,10 +407,8 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
         dofr[c] = *reinterpret_cast<bf16x8*>(&rd);
 #pragma unroll
         for (int j = 0; j < 8; j++) {
-          const int dd = 32 * c + 8 * hi + j;
-          const int off = ((c * 2 + (lo >> 3)) * 32 + (dd & 31)) * 8 + (lo & 7);
-          QT[wid][off] = rq[j];
-          dOT[wid][off] = rd[j];
+          QT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rq[j];
+          dOT[wid][swz_idx<32>(32 * c + 8 * hi + j, lo * 2)] = rd[j];
         }
       }
     }
@@ -467,9 +467,10 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_kernel(const short* __restrict_
       bf16x8 sA = *reinterpret_cast<bf16x8*>(&rs);
 #pragma unroll
       for (int c = 0; c < NC; c++) {
-        const int boff = ((c * 2 + hi5) * 32 + lo5) * 8;
-        s16x8 rdo = *reinterpret_cast<const s16x8*>(&dOT[wid][boff]);
-        s16x8 rqt = *reinterpret_cast<const s16x8*>(&QT[wid][boff]);
+        s16x8 rdo = *reinterpret_cast<const s16x8*>(
+            &dOT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
+        s16x8 rqt = *reinterpret_cast<const s16x8*>(
+            &QT[wid][swz_idx<32>(32 * c + lo5, 16 * hi5)]);
         acc_dv[c] = mfma32x32x16(pA, *reinterpret_cast<bf16x8*>(&rdo), acc_dv[c]);
         acc_dk[c] = mfma32x32x16(sA, *reinterpret_cast<bf16x8*>(&rqt), acc_dk[c]);
       }
