@@ -72,8 +72,8 @@ DEVINL unsigned pack2bf(float lo, float hi) {
 // in registers, and PV is ALSO computed transposed — O^T = mfma(V^T, P^T)
 // — which lands each lane's O accumulator in its OWN q-row column, so the
 // rescale and the 1/s epilogue are lane-local too.
-template <int D, bool CAUSAL>
-__global__ __launch_bounds__(512, 1)
+template <int D, bool CAUSAL, int NW = 8>
+__global__ __launch_bounds__(NW * 64, NW == 8 ? 1 : 2)
 void fa_fwd_kernel8(const short* __restrict__ q,
                     const short* __restrict__ k,
                     const short* __restrict__ v,
@@ -82,7 +82,8 @@ void fa_fwd_kernel8(const short* __restrict__ q,
                     float* __restrict__ lse,
                     int Hq, int Hkv, float scale) {
   constexpr int QBLK = 32;            // q rows per wave
-  constexpr int QTILE = 8 * QBLK;     // 256 per workgroup
+  constexpr int QTILE = NW * QBLK;    // rows per workgroup
+  constexpr int NT = NW * 64;         // threads
   constexpr int KVBLK = 64;
   constexpr int NDS = D / 16;         // d-slots per QK^T chain
   constexpr int NDT = D / 32;         // O d-tiles
@@ -129,13 +130,13 @@ void fa_fwd_kernel8(const short* __restrict__ q,
   // async-STAGE split staging (issue-early / write-late): 16 B per thread
   // per slice, 512 threads.
   constexpr int CHUNKS = KVBLK * D / 8;          // 16-B chunks per K (or V) tile
-  constexpr int NSLICE = (CHUNKS + 511) / 512;
+  constexpr int NSLICE = (CHUNKS + NT - 1) / NT;
   s16x8 pk_[NSLICE], pv_[NSLICE];
   auto issue_tile_loads = [&](int kv0) {
 #pragma unroll
     for (int sl = 0; sl < NSLICE; sl++) {
-      const int idx = threadIdx.x + sl * 512;
-      if (CHUNKS < 512 && idx >= CHUNKS) continue;
+      const int idx = threadIdx.x + sl * NT;
+      if (CHUNKS < NT && idx >= CHUNKS) continue;
       const int tok = idx / (D / 8);
       const int d0 = (idx % (D / 8)) * 8;
       const int kvi = min(kv0 + tok, len - 1);
@@ -147,8 +148,8 @@ void fa_fwd_kernel8(const short* __restrict__ q,
   auto write_tile_lds = [&](int buf) {
 #pragma unroll
     for (int sl = 0; sl < NSLICE; sl++) {
-      const int idx = threadIdx.x + sl * 512;
-      if (CHUNKS < 512 && idx >= CHUNKS) continue;
+      const int idx = threadIdx.x + sl * NT;
+      if (CHUNKS < NT && idx >= CHUNKS) continue;
       const int tok = idx / (D / 8);
       const int d0 = (idx % (D / 8)) * 8;
       *reinterpret_cast<s16x8*>(&Kt[buf][swz_idx<2 * D>(tok, d0 * 2)]) = pk_[sl];
@@ -524,19 +525,26 @@ std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
   if (T == 0) return {o, lse};
-  const int qtiles = (int)((max_seqlen + 255) / 256);
-  dim3 grid(qtiles, B, Hq), block(512);
+  static const int NW = [] {
+    const char* e = getenv("NANORLHF_FA_NW");
+    return (e && atoi(e) == 8) ? 8 : 4;   // 4-wave default (A/B-measured)
+  }();
+  const int qtile = NW * 32;
+  const int qtiles = (int)((max_seqlen + qtile - 1) / qtile);
+  dim3 grid(qtiles, B, Hq), block(NW * 64);
   auto stream = at::hip::getCurrentHIPStream();
-#define FWD_LAUNCH(DD, CC)                                                     \
-  hipLaunchKernelGGL((fa_fwd_kernel8<DD, CC>), grid, block, 0, stream,         \
+#define FWD_LAUNCH(DD, CC, NWW)                                                \
+  hipLaunchKernelGGL((fa_fwd_kernel8<DD, CC, NWW>), grid, block, 0, stream,    \
                      (const short*)q.data_ptr(), (const short*)k.data_ptr(),   \
                      (const short*)v.data_ptr(), cu_seqlens.data_ptr<int>(),   \
                      (short*)o.data_ptr(), lse.data_ptr<float>(), Hq, Hkv,     \
                      (float)scale)
-  if (D == 128) { if (causal) FWD_LAUNCH(128, true); else FWD_LAUNCH(128, false); }
-  else if (D == 64) { if (causal) FWD_LAUNCH(64, true); else FWD_LAUNCH(64, false); }
-  else if (D == 32) { if (causal) FWD_LAUNCH(32, true); else FWD_LAUNCH(32, false); }
+#define FWD_D(DD, CC) do { if (NW == 8) FWD_LAUNCH(DD, CC, 8); else FWD_LAUNCH(DD, CC, 4); } while (0)
+  if (D == 128) { if (causal) FWD_D(128, true); else FWD_D(128, false); }
+  else if (D == 64) { if (causal) FWD_D(64, true); else FWD_D(64, false); }
+  else if (D == 32) { if (causal) FWD_D(32, true); else FWD_D(32, false); }
   else TORCH_CHECK(false, "unsupported head_dim ", D);
+#undef FWD_D
 #undef FWD_LAUNCH
   HIP_CHECK_LAST();
   return {o, lse};
