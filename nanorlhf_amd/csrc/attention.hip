@@ -525,9 +525,12 @@ std::vector<torch::Tensor> fa_fwd_varlen(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({T, Hq}, q.options().dtype(torch::kFloat32));
   if (T == 0) return {o, lse};
+  // 8-wave default: the 4-wave/2-block variant A/B-measured SLOWER at
+  // every shape (244 vs 287 TF at the scoring bucket) — cross-block
+  // overlap does not make up for the extra per-thread staging work.
   static const int NW = [] {
     const char* e = getenv("NANORLHF_FA_NW");
-    return (e && atoi(e) == 8) ? 8 : 4;   // 4-wave default (A/B-measured)
+    return (e && atoi(e) == 4) ? 4 : 8;
   }();
   const int qtile = NW * 32;
   const int qtiles = (int)((max_seqlen + qtile - 1) / qtile);
