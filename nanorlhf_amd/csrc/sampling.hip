@@ -34,7 +34,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
                                    const long* __restrict__ step_ptr) {
   const unsigned long long step = step_ptr ? (unsigned long long)*step_ptr : step_imm;
   __shared__ float hist[NBINS];
-  __shared__ float sm[SBLOCK / 64], ss[SBLOCK / 64];
+  __shared__ float sm[SBLOCK / 64];
   __shared__ float wm_[SBLOCK / 64];
   __shared__ int wa_[SBLOCK / 64];
   __shared__ int argmax_sh;
@@ -45,8 +45,12 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int nvec = V / 8;
 
-  // ---- pass 1: fused online (m, s) + argmax ----------------------------
-  float m = -INFINITY, s = 0.f, lmax = -INFINITY;
+  // ---- pass 1: plain row max + argmax ----------------------------------
+  // (the round-1 online (m,s) update carried m,s ACROSS loop iterations —
+  // a serial exp/fma dependency chain per wave; plain fmax has none, and
+  // the exp-sum folds into the histogram pass below, which also drops one
+  // full read of the row)
+  float lmax = -INFINITY;
   int am = 0;
   const float invT = temperature > 0.f ? 1.f / temperature : 1.f;
   for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
@@ -55,68 +59,63 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     for (int j = 0; j < 8; j++) {
       const float l = bf2f(v[j]);
       if (l > lmax) { lmax = l; am = i * 8 + j; }
-      const float u = l * invT;
-      if (u > m) { s = s * __expf(m - u) + 1.f; m = u; }
-      else s += __expf(u - m);
     }
   }
   for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
     const float l = bf2f(lr[i]);
     if (l > lmax) { lmax = l; am = i; }
-    const float u = l * invT;
-    if (u > m) { s = s * __expf(m - u) + 1.f; m = u; }
-    else s += __expf(u - m);
   }
-  {  // merge (m, s) and argmax across the block
+  {  // merge argmax across the block
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) {
-      float m2 = __shfl_xor(m, off), s2 = __shfl_xor(s, off);
       float mx2 = __shfl_xor(lmax, off);
       int a2 = __shfl_xor(am, off);
-      const float mn = fmaxf(m, m2);
-      // -inf guards: threads with no elements (V < 8*SBLOCK) hold m = -inf
-      // and exp(-inf - -inf) would poison the merge with NaN
-      const float c1 = (m == -INFINITY) ? 0.f : __expf(m - mn);
-      const float c2 = (m2 == -INFINITY) ? 0.f : __expf(m2 - mn);
-      s = s * c1 + s2 * c2;
-      m = mn;
       if (mx2 > lmax || (mx2 == lmax && a2 < am)) { lmax = mx2; am = a2; }
     }
-    if (lane == 0) { sm[wid] = m; ss[wid] = s; wm_[wid] = lmax; wa_[wid] = am; }
+    if (lane == 0) { wm_[wid] = lmax; wa_[wid] = am; }
     __syncthreads();
     if (threadIdx.x == 0) {
-      float M = sm[0], S = ss[0], MX = wm_[0];
+      float MX = wm_[0];
       int A = wa_[0];
-      for (int w = 1; w < SBLOCK / 64; w++) {
-        const float mn = fmaxf(M, sm[w]);
-        const float c1 = (M == -INFINITY) ? 0.f : __expf(M - mn);
-        const float c2 = (sm[w] == -INFINITY) ? 0.f : __expf(sm[w] - mn);
-        S = S * c1 + ss[w] * c2;
-        M = mn;
+      for (int w = 1; w < SBLOCK / 64; w++)
         if (wm_[w] > MX || (wm_[w] == MX && wa_[w] < A)) { MX = wm_[w]; A = wa_[w]; }
-      }
-      sm[0] = M; ss[0] = S; argmax_sh = A;
+      wm_[0] = MX; argmax_sh = A;
     }
     __syncthreads();
   }
+  const float M = wm_[0] * invT;        // max of temperature-scaled logits
   if (temperature == 0.f) {
-    if (threadIdx.x == 0) {
+    // greedy: one extra pass only to report the T=1-softmax logprob
+    if (out_lp) {
+      float se = 0.f;
+      const float M1 = wm_[0];
+      for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
+#pragma unroll
+        for (int j = 0; j < 8; j++) se += __expf(bf2f(v[j]) - M1);
+      }
+      for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK)
+        se += __expf(bf2f(lr[i]) - M1);
+      se = block_sum<SBLOCK>(se, sm);
+      if (threadIdx.x == 0) {
+        out[row] = argmax_sh;
+        out_lp[row] = bf2f(lr[argmax_sh]) - (M1 + __logf(se));
+      }
+    } else if (threadIdx.x == 0) {
       out[row] = argmax_sh;
-      // greedy logprob reported under the T=1 softmax (vLLM convention)
-      if (out_lp) out_lp[row] = bf2f(lr[argmax_sh]) - (sm[0] + __logf(ss[0]));
     }
     return;
   }
-  const float M = sm[0];          // max of temperature-scaled logits
-  const float invS = 1.f / ss[0];
 
   // ---- pass 2: coarse histogram of prob mass (wave-privatized: 4 waves
   // hammering one 1024-bin histogram serialize on LDS atomics; a private
   // histogram per wave merged once removes the cross-wave conflicts) ----
   __shared__ float hist4[SBLOCK / 64][NBINS];
+  __shared__ float S_sh;
   for (int w = 0; w < SBLOCK / 64; w++)
     for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist4[w][i] = 0.f;
   __syncthreads();
+  float sacc = 0.f;   // exp-sum accumulated in the SAME pass (unnormalized)
   for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
     s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
 #pragma unroll
@@ -124,10 +123,24 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       const float u = bf2f(v[j]) * invT - M;
       int b = (int)((u + URANGE) * (NBINS / URANGE));
       b = max(0, min(NBINS - 1, b));
-      atomicAdd(&hist4[wid][b], __expf(u) * invS);
+      const float e = __expf(u);
+      sacc += e;
+      atomicAdd(&hist4[wid][b], e);
     }
   }
+  for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
+    const float u = bf2f(lr[i]) * invT - M;
+    int b = (int)((u + URANGE) * (NBINS / URANGE));
+    b = max(0, min(NBINS - 1, b));
+    const float e = __expf(u);
+    sacc += e;
+    atomicAdd(&hist4[wid][b], e);
+  }
+  sacc = block_sum<SBLOCK>(sacc, sm);
+  if (threadIdx.x == 0) S_sh = sacc;
   __syncthreads();
+  const float S = S_sh;               // histogram/masses stay UNNORMALIZED:
+  const float ptarget = top_p * S;    // compare against top_p * S instead
   for (int i = threadIdx.x; i < NBINS; i += SBLOCK)
     hist[i] = hist4[0][i] + hist4[1][i] + hist4[2][i] + hist4[3][i];
   __syncthreads();
@@ -139,7 +152,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     mass_above_sh = 0.f;
     for (int b = NBINS - 1; b >= 0; b--) {
       const float nacc = acc + hist[b];
-      if (nacc >= top_p || b == 0) { bstar = b; mass_above_sh = acc; break; }
+      if (nacc >= ptarget || b == 0) { bstar = b; mass_above_sh = acc; break; }
       acc = nacc;
     }
     bin_star_sh = bstar;
@@ -157,7 +170,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   __shared__ int skip_refine_sh;
   if (threadIdx.x == 0) {
     const float bin_mass = hist[bstar];
-    skip_refine_sh = (bin_mass <= 0.02f) ? 1 : 0;
+    skip_refine_sh = (bin_mass <= 0.02f * S) ? 1 : 0;
     if (skip_refine_sh) { u_thresh_sh = bin_lo; found_sh = -1; }
   }
   __syncthreads();
@@ -177,7 +190,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       if (cb == bstar) {
         int b = (int)((u - bin_lo) * sub_scale);
         b = max(0, min(NBINS - 1, b));
-        atomicAdd(&hist[b], __expf(u) * invS);
+        atomicAdd(&hist[b], __expf(u));
       }
     }
   }
@@ -187,7 +200,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     float thresh = bin_lo;
     for (int b = NBINS - 1; b >= 0; b--) {
       acc += hist[b];
-      if (acc >= top_p || b == 0) {
+      if (acc >= ptarget || b == 0) {
         thresh = bin_lo + (float)b * (bin_hi - bin_lo) / NBINS;
         break;
       }
@@ -206,12 +219,12 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
 #pragma unroll
     for (int j = 0; j < 8; j++) {
       const float u = bf2f(v[j]) * invT - M;
-      if (u >= u_thresh) own += __expf(u) * invS;
+      if (u >= u_thresh) own += __expf(u);
     }
   }
   for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
     const float u = bf2f(lr[i]) * invT - M;
-    if (u >= u_thresh) own += __expf(u) * invS;
+    if (u >= u_thresh) own += __expf(u);
   }
   float incl = warp_incl_scan(own, lane);
   if (lane == 63) wsum[wid] = incl;
@@ -239,7 +252,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
       for (int j = 0; j < 8; j++) {
         const float u = bf2f(v[j]) * invT - M;
         if (u >= u_thresh) {
-          acc += __expf(u) * invS;
+          acc += __expf(u);
           if (target < acc && found < 0) found = i * 8 + j;
         }
       }
@@ -247,7 +260,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
       const float u = bf2f(lr[i]) * invT - M;
       if (u >= u_thresh) {
-        acc += __expf(u) * invS;
+        acc += __expf(u);
         if (target < acc && found < 0) found = i;
       }
     }
@@ -259,7 +272,7 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
     out[row] = f;
     // logprob of the chosen token under the temperature-scaled softmax
     // (matches the scoring pass's logits/temperature quirk, grpo_trainer.py:547)
-    if (out_lp) out_lp[row] = bf2f(lr[f]) * invT - (M + __logf(ss[0]));
+    if (out_lp) out_lp[row] = bf2f(lr[f]) * invT - (M + __logf(S));
   }
 }
 
