@@ -33,7 +33,6 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
                                    unsigned long long step_imm,
                                    const long* __restrict__ step_ptr) {
   const unsigned long long step = step_ptr ? (unsigned long long)*step_ptr : step_imm;
-  __shared__ float hist[NBINS];
   __shared__ float sm[SBLOCK / 64];
   __shared__ float wm_[SBLOCK / 64];
   __shared__ int wa_[SBLOCK / 64];
@@ -110,116 +109,134 @@ __global__ void sample_topp_kernel(const short* __restrict__ logits,
   // ---- pass 2: coarse histogram of prob mass (wave-privatized: 4 waves
   // hammering one 1024-bin histogram serialize on LDS atomics; a private
   // histogram per wave merged once removes the cross-wave conflicts) ----
-  // The histogram atomics were 75% of the whole kernel (ablation
-  // tools/topp_ablate.hip: 0.41 ms streaming → 1.69 ms with atomics):
-  // exp mass clusters into few bins, so up to 64 lanes of a wave
-  // serialize on one ds_add_f32.  Spread each bin over 8 lane-keyed
-  // slots (lane&7) — same-bin updates from one wave land on 8 banks.
-  __shared__ float histS[NBINS * 8];
-  __shared__ float S_sh;
-  for (int i = threadIdx.x; i < NBINS * 8; i += SBLOCK) histS[i] = 0.f;
-  __syncthreads();
-  const int lkey = threadIdx.x & 7;
-  float sacc = 0.f;   // exp-sum accumulated in the SAME pass (unnormalized)
-  for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
-    s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
+  // Threshold search WITHOUT LDS atomics.  Ablation (tools/topp_ablate.hip)
+  // showed the 1024-bin atomicAdd histogram was 75% of the kernel — LDS
+  // f32 atomics are throughput-bound regardless of conflict spreading
+  // (lane-keyed slots measured no better).  Instead: a 32-bin REGISTER
+  // histogram per thread (no atomics), block-reduced through LDS once,
+  // then the boundary bin is refined by re-binning only its interior on
+  // additional streaming passes (each pass is pure reads + VALU at ~6 TB/s).
+  __shared__ float Hist32[32 * (SBLOCK / 64)];
+  __shared__ float S_sh, lo_sh, span_sh, above_sh;
+  __shared__ int done_sh;
+  float sacc = 0.f;
+  {
+    float hb[32];
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      const float u = bf2f(v[j]) * invT - M;
-      int b = (int)((u + URANGE) * (NBINS / URANGE));
-      b = max(0, min(NBINS - 1, b));
+    for (int b = 0; b < 32; b++) hb[b] = 0.f;
+    for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float u = bf2f(v[j]) * invT - M;
+        const float e = __expf(u);
+        sacc += e;
+        int b = (int)(u + URANGE);
+        b = max(0, min(31, b));
+        hb[b] += e;
+      }
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
+      const float u = bf2f(lr[i]) * invT - M;
       const float e = __expf(u);
       sacc += e;
-      atomicAdd(&histS[b * 8 + lkey], e);
+      int b = (int)(u + URANGE);
+      b = max(0, min(31, b));
+      hb[b] += e;
     }
-  }
-  for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
-    const float u = bf2f(lr[i]) * invT - M;
-    int b = (int)((u + URANGE) * (NBINS / URANGE));
-    b = max(0, min(NBINS - 1, b));
-    const float e = __expf(u);
-    sacc += e;
-    atomicAdd(&histS[b * 8 + lkey], e);
-  }
-  sacc = block_sum<SBLOCK>(sacc, sm);
-  if (threadIdx.x == 0) S_sh = sacc;
-  __syncthreads();
-  const float S = S_sh;               // histogram/masses stay UNNORMALIZED:
-  const float ptarget = top_p * S;    // compare against top_p * S instead
-  for (int i = threadIdx.x; i < NBINS; i += SBLOCK) {
-    float h = 0.f;
+    // wave-reduce each bin, then one slot per (bin, wave) in LDS
 #pragma unroll
-    for (int k2 = 0; k2 < 8; k2++) h += histS[i * 8 + k2];
-    hist[i] = h;
+    for (int b = 0; b < 32; b++) {
+      float x = hb[b];
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off);
+      if (lane == 0) Hist32[b * (SBLOCK / 64) + wid] = x;
+    }
+    sacc = block_sum<SBLOCK>(sacc, sm);
+    if (threadIdx.x == 0) S_sh = sacc;
+    __syncthreads();
   }
-  __syncthreads();
-  __shared__ float u_thresh_sh, mass_above_sh;
-  __shared__ int bin_star_sh;
+  const float S = S_sh;
+  const float ptarget = top_p * S;
+  // coarse bin select (bins are 1 nat wide, u in [-URANGE, 0])
   if (threadIdx.x == 0) {
     float acc = 0.f;
     int bstar = 0;
-    mass_above_sh = 0.f;
-    for (int b = NBINS - 1; b >= 0; b--) {
-      const float nacc = acc + hist[b];
-      if (nacc >= ptarget || b == 0) { bstar = b; mass_above_sh = acc; break; }
+    float above = 0.f;
+    for (int b = 31; b >= 0; b--) {
+      float h = 0.f;
+      for (int w = 0; w < SBLOCK / 64; w++) h += Hist32[b * (SBLOCK / 64) + w];
+      const float nacc = acc + h;
+      if (nacc >= ptarget || b == 0) { bstar = b; above = acc; break; }
       acc = nacc;
     }
-    bin_star_sh = bstar;
-  }
-  __syncthreads();
-  const int bstar = bin_star_sh;
-  const float bin_lo = (float)bstar * (URANGE / NBINS) - URANGE;
-  const float bin_hi = bin_lo + (URANGE / NBINS);
-
-  // refinement is only worth a full extra pass when the boundary bin
-  // holds real mass: keeping the WHOLE bin inflates the nucleus by at
-  // most hist[bstar], so when that is under a 2% slack just take
-  // u_thresh = bin_lo (the coarse histogram already places it within
-  // 1/32 nat) and skip pass 3 entirely (~20% of the kernel's traffic)
-  __shared__ int skip_refine_sh;
-  if (threadIdx.x == 0) {
-    const float bin_mass = hist[bstar];
-    skip_refine_sh = (bin_mass <= 0.02f * S) ? 1 : 0;
-    if (skip_refine_sh) { u_thresh_sh = bin_lo; found_sh = -1; }
-  }
-  __syncthreads();
-  if (!skip_refine_sh) {
-  // ---- pass 3: refinement histogram inside bin* (membership by the SAME
-  // coarse binning, so boundary values like u == 0 stay consistent) ------
-  for (int i = threadIdx.x; i < NBINS; i += SBLOCK) hist[i] = 0.f;
-  __syncthreads();
-  const float sub_scale = NBINS / (bin_hi - bin_lo);
-  for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
-    s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
-#pragma unroll
-    for (int j = 0; j < 8; j++) {
-      const float u = bf2f(v[j]) * invT - M;
-      int cb = (int)((u + URANGE) * (NBINS / URANGE));
-      cb = max(0, min(NBINS - 1, cb));
-      if (cb == bstar) {
-        int b = (int)((u - bin_lo) * sub_scale);
-        b = max(0, min(NBINS - 1, b));
-        atomicAdd(&hist[b], __expf(u));
-      }
-    }
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float acc = mass_above_sh;
-    float thresh = bin_lo;
-    for (int b = NBINS - 1; b >= 0; b--) {
-      acc += hist[b];
-      if (acc >= ptarget || b == 0) {
-        thresh = bin_lo + (float)b * (bin_hi - bin_lo) / NBINS;
-        break;
-      }
-    }
-    u_thresh_sh = thresh;
+    lo_sh = (float)bstar - URANGE;
+    span_sh = 1.0f;
+    above_sh = above;
+    done_sh = 0;
     found_sh = -1;
   }
   __syncthreads();
-  }  // !skip_refine
-  const float u_thresh = u_thresh_sh;
+  // refine the boundary bin: re-bin its interior into 32 sub-bins per pass
+  // until the bin is narrow (1/1024 nat) or holds negligible mass
+  for (int it = 0; it < 2 && !done_sh; it++) {
+    const float lo = lo_sh, span = span_sh;
+    float hb[32];
+#pragma unroll
+    for (int b = 0; b < 32; b++) hb[b] = 0.f;
+    const float sub = 32.0f / span;
+    for (int i = threadIdx.x; i < nvec; i += SBLOCK) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(lr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float u = bf2f(v[j]) * invT - M;
+        if (u >= lo && u < lo + span) {
+          int b = (int)((u - lo) * sub);
+          b = max(0, min(31, b));
+          hb[b] += __expf(u);
+        }
+      }
+    }
+    for (int i = nvec * 8 + threadIdx.x; i < V; i += SBLOCK) {
+      const float u = bf2f(lr[i]) * invT - M;
+      if (u >= lo && u < lo + span) {
+        int b = (int)((u - lo) * sub);
+        b = max(0, min(31, b));
+        hb[b] += __expf(u);
+      }
+    }
+#pragma unroll
+    for (int b = 0; b < 32; b++) {
+      float x = hb[b];
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off);
+      if (lane == 0) Hist32[b * (SBLOCK / 64) + wid] = x;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float acc = above_sh;
+      int bstar = 0;
+      float above = above_sh;
+      for (int b = 31; b >= 0; b--) {
+        float h = 0.f;
+        for (int w = 0; w < SBLOCK / 64; w++) h += Hist32[b * (SBLOCK / 64) + w];
+        const float nacc = acc + h;
+        if (nacc >= ptarget || b == 0) { bstar = b; above = acc; break; }
+        acc = nacc;
+      }
+      lo_sh = lo + (float)bstar * (span / 32.0f);
+      span_sh = span / 32.0f;
+      above_sh = above;
+      // stop when the boundary sliver is negligible (over-keeping the
+      // whole sliver changes the nucleus by < 2% of S)
+      float bmass = 0.f;
+      for (int w = 0; w < SBLOCK / 64; w++)
+        bmass += Hist32[bstar * (SBLOCK / 64) + w];
+      if (bmass <= 0.02f * S) done_sh = 1;
+    }
+    __syncthreads();
+  }
+  const float u_thresh = lo_sh;
 
   // ---- pass 4: per-thread kept mass + block scan + owner walk ----------
   float own = 0.f;
