@@ -153,9 +153,14 @@ void fa_fwd_kernel8(const short* __restrict__ q,
       const int tok = idx / (D / 8);
       const int d0 = (idx % (D / 8)) * 8;
       *reinterpret_cast<s16x8*>(&Kt[buf][swz_idx<2 * D>(tok, d0 * 2)]) = pk_[sl];
+      // V fragment-major: slot (kv>>3, d) = V^T[d][8 contiguous kv] — PV
+      // reads walk consecutive slots per lane (conflict-free).  Measured
+      // wall-neutral vs the swizzled [d][kv] row image (288 vs 287 TF —
+      // the fwd is barrier/imbalance-bound, not LDS-bound); kept for the
+      // simpler addressing
 #pragma unroll
       for (int j = 0; j < 8; j++)
-        Vt[buf][swz_idx<2 * KVBLK>(d0 + j, tok * 2)] = pv_[sl][j];
+        Vt[buf][(((tok >> 3) * D) + d0 + j) * 8 + (tok & 7)] = pv_[sl][j];
     }
   };
 
@@ -264,7 +269,7 @@ void fa_fwd_kernel8(const short* __restrict__ q,
 #pragma unroll
       for (int ks = 0; ks < 4; ks++) {
         s16x8 rv = *reinterpret_cast<const s16x8*>(
-            &Vt[buf][swz_idx<2 * KVBLK>(t * 32 + col, (16 * ks + 8 * hi) * 2)]);
+            &Vt[buf][((2 * ks + hi) * D + t * 32 + col) * 8]);
         acc_o[t] = mfma32x32x16(*reinterpret_cast<bf16x8*>(&rv), pa[ks], acc_o[t]);
       }
     }
