@@ -210,3 +210,26 @@ def save_hf_checkpoint(model: CausalLM, path: str) -> None:
         json.dump(hf_cfg, f, indent=2)
     state = {k: v.contiguous() for k, v in unfuse_qwen2_state(model).items()}
     save_file(state, os.path.join(path, "model.safetensors"))
+
+
+def export_merged_hf(model: CausalLM, path: str) -> None:
+    """Export a (possibly LoRA-wrapped) trained policy as a plain HF Qwen2
+    checkpoint with adapters merged into the base weights — the artifact a
+    user deploys or hands to any HF-ecosystem tool.  (The reference writes
+    merged checkpoints to disk every update to feed vLLM,
+    grpo_trainer.py:131-140; here merging is only ever done on export.)"""
+    import copy
+
+    from .lora import LoRALinear
+
+    model = copy.deepcopy(model).cpu()
+    for name, module in list(model.named_modules()):
+        for child_name, child in list(module.named_children()):
+            if isinstance(child, LoRALinear):
+                with torch.no_grad():
+                    merged = (child.base.weight.float()
+                              + (child.lora_B.float() @ child.lora_A.float())
+                              * child.scaling).to(child.base.weight.dtype)
+                    child.base.weight.copy_(merged)
+                setattr(module, child_name, child.base)
+    save_hf_checkpoint(model, path)

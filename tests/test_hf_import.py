@@ -343,3 +343,36 @@ def test_r1_real_data_seam_end_to_end(tmp_path, tiny_tok):
     # checkpoint written by trainer.save()
     assert any(p.name.startswith("checkpoint-")
                for p in (tmp_path / "out").iterdir())
+
+
+def test_export_merged_hf_roundtrip(tmp_path):
+    """Trained-model deploy path: LoRA-wrapped model → merged HF checkpoint
+    → reload → logits equal the live merged forward."""
+    from nanorlhf_amd.models.lora import LoraConfig, apply_lora
+    torch.manual_seed(2)
+    cfg = ModelConfig(vocab_size=256, hidden_size=64, num_layers=2, num_heads=4,
+                      num_kv_heads=2, head_dim=16, intermediate_size=128,
+                      rope_theta=1e4, max_position=256, dtype="float32",
+                      tie_word_embeddings=True)
+    m = CausalLM(cfg)
+    apply_lora(m, LoraConfig(r=4, alpha=8))
+    # give the adapters non-zero effect
+    with torch.no_grad():
+        for n, p in m.named_parameters():
+            if "lora_B" in n:
+                p.normal_(0, 0.05)
+    seqs = [torch.randint(0, 256, (9,))]
+    ids, cu, mx, pos = pack_sequences(seqs)
+    ctx = CausalLM.train_ctx(cu, mx, pos)
+    with torch.no_grad():
+        want = m.logits(m(ids, ctx))
+    out_dir = str(tmp_path / "deploy")
+    hf_import.export_merged_hf(m, out_dir)
+    # exported model is plain (no adapters) and reproduces the merged logits
+    m2 = hf_import.load_pretrained(out_dir).eval()
+    assert not any("lora" in k for k in m2.state_dict())
+    with torch.no_grad():
+        got = m2.logits(m2(ids, ctx))
+    assert torch.allclose(got, want, atol=1e-4), float((got - want).abs().max())
+    # the ORIGINAL model is untouched (still adapter-wrapped)
+    assert any("lora_A" in k for k, _ in m.named_parameters())
