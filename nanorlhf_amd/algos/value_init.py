@@ -38,7 +38,7 @@ def finetune_value_model(trainer, num_prompts: int = 500, epochs: int = 8,
             while row and row[-1] == cfg.pad_token_id:
                 row.pop()
         responses.append(row)
-    scores = trainer.reward_fn([list(p) + r for p, r in zip(prompts, responses)]).float()
+    scores = trainer._call_reward([list(p) for p in prompts], responses)
     lp, ref_lp, _, mask, _ = trainer.score_rows(prompts, responses, with_ref=True)
     eos_idx = mask.sum(1).long() - 1
     rewards = F.kl_shaped_rewards(scores.to(device), lp, ref_lp, mask, eos_idx, cfg.kl_coef)

@@ -133,3 +133,35 @@ def test_parse_structured_kinds():
     assert parse_structured(r"\begin{pmatrix}1\\2\end{pmatrix}").kind == "matrix"
     assert parse_structured("42") is None
     assert parse_structured("(x+1)*(x+2)") is None
+
+
+# ---- property tests ------------------------------------------------------
+
+def test_reflexive_on_numbers():
+    import random
+    rng = random.Random(0)
+    for _ in range(40):
+        x = rng.choice([str(rng.randint(-10**6, 10**6)),
+                        f"{rng.uniform(-100, 100):.4f}",
+                        f"{rng.randint(1, 99)}/{rng.randint(1, 99)}"])
+        assert answers_equal(x, x), x
+
+
+def test_fraction_identities():
+    import random
+    rng = random.Random(1)
+    for _ in range(25):
+        a, b = rng.randint(1, 50), rng.randint(1, 50)
+        k = rng.randint(2, 5)
+        assert answers_equal(rf"\frac{{{a}}}{{{b}}}",
+                             rf"\frac{{{a * k}}}{{{b * k}}}"), (a, b, k)
+        assert answers_equal(rf"\frac{{{a}}}{{{b}}}", f"{a / b:.6f}"), (a, b)
+
+
+def test_negative_pairs_random():
+    import random
+    rng = random.Random(2)
+    for _ in range(25):
+        a = rng.randint(0, 10**5)
+        b = a + rng.randint(1, 9) * max(1, a // 50 + 1)
+        assert not answers_equal(str(a), str(b)), (a, b)
