@@ -230,3 +230,23 @@ def test_7b_geometry_small_depth():
     (-lp.mean()).backward()
     g = m.model.layers[0].self_attn.qkv_proj.weight.grad
     assert g is not None and torch.isfinite(g.float()).all()
+
+
+def test_deberta_gpu_matches_cpu():
+    """DeBERTa-v3 RM on GPU bf16 vs the CPU fp32 reference (the real-RM
+    scoring path exercised on device)."""
+    from nanorlhf_amd.models.deberta import DebertaConfig, DebertaV3Reward
+    torch.manual_seed(0)
+    cfg = DebertaConfig(vocab_size=2048, hidden_size=128, num_layers=4,
+                        num_heads=8, intermediate_size=256,
+                        max_position_embeddings=128, position_buckets=32,
+                        position_biased_input=False, num_labels=1)
+    m = DebertaV3Reward(cfg).eval()
+    ids = torch.randint(0, 2048, (4, 96))
+    mask = torch.ones(4, 96, dtype=torch.long)
+    mask[1, 60:] = 0
+    with torch.no_grad():
+        want = m(ids, mask)
+        got = m.to(DEV).to(torch.bfloat16)(ids.to(DEV), mask.to(DEV))
+    err = float((got.float().cpu() - want).abs().max() / (want.abs().max() + 1e-6))
+    assert err < 0.08, err
