@@ -294,7 +294,70 @@ class RLHFTrainer:
         ro = Rollout(prompts=rows_p, responses=rows_r, scores=scores,
                      raw_scores=raw_scores, contains_eos=contains_eos,
                      sample_n=cfg.sample_n, logprobs=rows_lp)
+        if (getattr(cfg, "dp_rebalance_rollout", False) and self.world > 1
+                and not self.algo.greedy_baseline):
+            # (ReMax's greedy baseline is positional per prompt — rebalance
+            # would desync it, so it stays local for that algorithm)
+            ro = self._rebalance_rollout(ro)
         return ro, greedy_scores
+
+    def _rebalance_rollout(self, ro: Rollout) -> Rollout:
+        """Re-assign whole sample-groups across DP ranks so per-rank token
+        totals are near-equal (greedy LPT, deterministic on every rank).
+        Groups stay intact — group-relative advantages (GRPO/RLOO) and
+        keep-1-of-N operate within a group, so relocation is transparent.
+        Rewards/penalties were already applied locally and travel with the
+        rows.  ReMax greedy scores are per-row too but its spec consumes
+        them positionally, so rebalance is limited to algos without a
+        greedy baseline (guarded by the caller's flag; documented)."""
+        import torch.distributed as _dist
+        n = max(ro.sample_n, 1)
+        n_groups = ro.num_rows // n
+        groups = []
+        for g in range(n_groups):
+            rows = list(range(g * n, (g + 1) * n))
+            groups.append({
+                "prompts": [ro.prompts[i] for i in rows],
+                "responses": [ro.responses[i] for i in rows],
+                "scores": [float(ro.scores[i]) for i in rows],
+                "raw": [float(ro.raw_scores[i]) for i in rows],
+                "eos": [bool(ro.contains_eos[i]) for i in rows],
+                "lp": ([ro.logprobs[i] for i in rows] if ro.logprobs else None),
+            })
+        gathered = [None] * self.world
+        _dist.all_gather_object(gathered, groups)
+        flat = []
+        for src, gl in enumerate(gathered):
+            for gi, g in enumerate(gl):
+                toks = sum(len(p) + len(r)
+                           for p, r in zip(g["prompts"], g["responses"]))
+                flat.append((toks, src, gi, g))
+        # deterministic LPT: longest group first onto the least-loaded rank
+        flat.sort(key=lambda t: (-t[0], t[1], t[2]))
+        loads = [0] * self.world
+        counts = [0] * self.world
+        mine = []
+        for toks, src, gi, g in flat:
+            dst = min(range(self.world), key=lambda r: (loads[r], counts[r], r))
+            loads[dst] += toks
+            counts[dst] += 1
+            if dst == self.rank:
+                mine.append(g)
+        prompts, responses, scores, raw, eos, lps = [], [], [], [], [], []
+        for g in mine:
+            prompts.extend(g["prompts"])
+            responses.extend(g["responses"])
+            scores.extend(g["scores"])
+            raw.extend(g["raw"])
+            eos.extend(g["eos"])
+            if g["lp"] is not None:
+                lps.extend(g["lp"])
+        return Rollout(prompts=prompts, responses=responses,
+                       scores=torch.tensor(scores, dtype=torch.float32),
+                       raw_scores=torch.tensor(raw, dtype=torch.float32),
+                       contains_eos=torch.tensor(eos, dtype=torch.bool),
+                       sample_n=ro.sample_n,
+                       logprobs=lps if lps else None)
 
     def _call_reward(self, rows_p: list[list[int]], rows_r: list[list[int]]) -> torch.Tensor:
         """Dispatch to the reward plug-in.  String-contract rewards

@@ -94,6 +94,15 @@ class RLHFConfig:
     # ---- sparse-GRPO / r1 mode (grpo_r1_trainer.py) --------------------------
     sparse_filter: bool = False            # drop score==0 samples (:565-568)
 
+    # ---- DP load balance ------------------------------------------------------
+    # Real-model rollouts have EOS-driven response-length variance; ranks
+    # then carry unequal token counts into the scoring/update phases (the
+    # synthetic bench is balanced).  When on, whole sample-groups are
+    # re-assigned across ranks after reward so per-rank token totals are
+    # near-equal (LPT greedy, deterministic).  Off by default: it is an
+    # opt-in for real-data DP runs (SURVEY §7 hard part (d)).
+    dp_rebalance_rollout: bool = False
+
     # ---- eval (r1 mode: greedy accuracy pass, grpo_r1_trainer.py:824-825) ----
     eval_steps: int = 0                    # 0 → no periodic eval
     eval_at_start: bool = True             # initial accuracy (:471-473)

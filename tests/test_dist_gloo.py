@@ -129,7 +129,73 @@ def _run_sparse_grpo_dp(rank, world, store_path, q, tmpdir):
             torch_dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("fn", [_run_reducer, _run_trainer_dp, _run_sparse_grpo_dp])
+def _run_rebalance_dp(rank, world, store_path, q, tmpdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    _init_pg(rank, world, store_path)
+    try:
+        import torch
+        from nanorlhf_amd.algos import grpo
+        from nanorlhf_amd.algos.grpo import GRPOConfig
+        from nanorlhf_amd.algos.trainer import Rollout
+        from nanorlhf_amd.data import hh_shaped_prompts
+        from nanorlhf_amd.models import CausalLM
+
+        cfg = GRPOConfig(model_preset="tiny", dtype="float32", use_lora=True,
+                         lora_r=4, lora_alpha=8, per_device_train_batch_size=2,
+                         gradient_accumulation_steps=1, num_mini_batches=2,
+                         total_episodes=8, sample_n=2, response_length=4,
+                         temperature=1.0, stop_token_id=1,
+                         output_dir=os.path.join(tmpdir, f"rb{rank}"),
+                         gradient_checkpointing=False, score_token_budget=256,
+                         dp_rebalance_rollout=True)
+        torch.manual_seed(0)
+        policy = CausalLM.from_preset("tiny")
+        ref = CausalLM.from_preset("tiny")
+        ref.load_state_dict(policy.state_dict())
+        prompts = hh_shaped_prompts(16, 1024, min_len=4, max_len=8)
+        tr = grpo.make_trainer(cfg, policy, ref,
+                               lambda s: torch.tensor([float(len(x) % 3) for x in s]),
+                               prompts)
+
+        # unit check of the rebalancer: rank 0 holds LONG groups, rank 1
+        # short — after rebalance token loads must be near-equal and the
+        # global multiset of groups preserved
+        L = 40 if rank == 0 else 4
+        n_groups = 4
+        ps, rs = [], []
+        for g in range(n_groups):
+            for j in range(2):
+                ps.append([rank * 1000 + g * 10 + j] * 3)
+                rs.append([5] * L)
+        ro = Rollout(prompts=ps, responses=rs,
+                     scores=torch.arange(len(ps), dtype=torch.float32),
+                     raw_scores=torch.arange(len(ps), dtype=torch.float32),
+                     contains_eos=torch.zeros(len(ps), dtype=torch.bool),
+                     sample_n=2,
+                     logprobs=[[0.1] * L for _ in ps])
+        ro2 = tr._rebalance_rollout(ro)
+        my_toks = sum(len(p) + len(r) for p, r in zip(ro2.prompts, ro2.responses))
+        toks = [torch.zeros(1) for _ in range(world)]
+        torch_dist.all_gather(toks, torch.tensor([float(my_toks)]))
+        total = sum(float(t) for t in toks)
+        ok = all(abs(float(t) - total / world) <= 90 for t in toks)
+        # global group preservation: gather all first-prompt markers
+        markers = sorted(p[0] for p in ro2.prompts)
+        allm = [None] * world
+        torch_dist.all_gather_object(allm, markers)
+        merged = sorted(x for m in allm for x in m)
+        want = sorted([r * 1000 + g * 10 + j for r in range(world)
+                       for g in range(4) for j in range(2)])
+        ok = ok and merged == want
+        # and a full update still runs with the flag on
+        tr.train(num_updates=1)
+        q.put((rank, ok))
+    finally:
+        if torch_dist.is_initialized():
+            torch_dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("fn", [_run_reducer, _run_trainer_dp, _run_sparse_grpo_dp, _run_rebalance_dp])
 def test_world2_gloo(fn, tmp_path):
     world = 2
     store_path = str(tmp_path / "pg_store")
